@@ -1,0 +1,50 @@
+"""Stage-sharded checkpoint save/load.
+
+The reference library has no checkpointing; only its side script saves
+a state_dict pickle (scripts/DDP_PyTorch_MNIST.py:157-161, naming
+precedent `model_p{size}.pkl`).  BASELINE.json requires a checkpoint
+layout, so: each pipeline stage writes its own file, written only by
+DP rank 0 of that stage (every DP replica is identical — enforced by
+assert_sync).  Layout:
+
+    <dir>/meta.pt                      (world/dp/pp, sizes, step)
+    <dir>/stage_{s:02d}.pt             (f32 master params, in order)
+"""
+
+import os
+
+import torch
+
+
+def save_checkpoint(path, model, topo, step: int = 0, extra=None):
+    os.makedirs(path, exist_ok=True)
+    if topo.dp_rank == 0:
+        state = {
+            "params": [p.data.detach().cpu() for p in model.parameters()],
+            "stage_id": topo.stage_id,
+            "step": step,
+        }
+        torch.save(state, os.path.join(path, f"stage_{topo.stage_id:02d}.pt"))
+    if topo.rank == 0:
+        meta = {"dp": topo.dp, "pp": topo.pp, "step": step}
+        if extra:
+            meta.update(extra)
+        torch.save(meta, os.path.join(path, "meta.pt"))
+
+
+def load_checkpoint(path, model, topo):
+    f = os.path.join(path, f"stage_{topo.stage_id:02d}.pt")
+    state = torch.load(f, map_location="cpu", weights_only=False)
+    params = model.parameters()
+    assert len(params) == len(state["params"]), (
+        f"checkpoint has {len(state['params'])} tensors, "
+        f"model stage has {len(params)}"
+    )
+    for p, saved in zip(params, state["params"]):
+        assert p.data.shape == saved.shape, (p.data.shape, saved.shape)
+        p.data.copy_(saved.to(p.data.device))
+        p.sync_lp()
+    meta_f = os.path.join(path, "meta.pt")
+    meta = torch.load(meta_f, map_location="cpu", weights_only=False) \
+        if os.path.exists(meta_f) else {}
+    return meta

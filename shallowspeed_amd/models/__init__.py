@@ -1,0 +1,12 @@
+from .layers import (  # noqa: F401
+    MLP,
+    Linear,
+    Module,
+    Parameter,
+    ReLU,
+    Sequential,
+    Softmax,
+    SoftmaxMSE,
+    SoftmaxXent,
+)
+from .optimizer import SGD  # noqa: F401

@@ -1,0 +1,51 @@
+"""Stateless SGD.  Reference: shallowspeed/optimizer.py:4-13.
+
+CPU: p.data -= lr * p.grad per parameter.
+GPU: ONE fused multi-tensor HIP kernel launch updates every parameter
+of the stage: f32 master -= lr * f32 grad, and re-emits the bf16
+compute copy AND the transposed bf16 copy in the same pass (the
+transposed copy is what makes dgrad an NT GEMM).
+"""
+
+import torch
+
+from ..ops._ext import load_ext
+
+
+class SGD:
+    def __init__(self, parameters, lr: float):
+        self.params = [p for p in parameters if p.requires_grad]
+        self.lr = float(lr)
+        self._desc = None  # device-side descriptor table, built lazily
+
+    def _build_desc(self, ext):
+        # Static descriptor: [master_ptr, grad_ptr, lp_ptr, lpt_ptr,
+        # numel, cols] per tensor, int64, lives on device.  Pointers are
+        # stable because parameter storage is persistent.
+        rows = []
+        for p in self.params:
+            t = p.data
+            lp = p.lp
+            lpt = p.lp_t
+            cols = t.shape[1] if t.dim() == 2 else 0
+            rows.append([
+                t.data_ptr(), p.grad.data_ptr(),
+                lp.data_ptr() if lp is not None else 0,
+                lpt.data_ptr() if lpt is not None else 0,
+                t.numel(), cols,
+            ])
+        cpu = torch.tensor(rows, dtype=torch.int64)
+        self._desc = cpu.to(self.params[0].data.device)
+        self._desc_cpu = cpu
+
+    def step(self):
+        if not self.params:
+            return
+        if self.params[0].data.is_cuda:
+            ext = load_ext(required=True)
+            if self._desc is None:
+                self._build_desc(ext)
+            ext.sgd_multi(self._desc, self.lr)
+        else:
+            for p in self.params:
+                p.data -= self.lr * p.grad

@@ -1,0 +1,184 @@
+"""Stateless math ops, dual-backend.
+
+Reference: shallowspeed/functional.py:4-44 — 8 pure NumPy functions
+(linear/relu/softmax/mse + grads).  Here every op has
+
+  * a CPU reference implementation in plain torch float32 — used by
+    host-only tests and as the numerics oracle for the HIP kernels, and
+  * a GPU implementation dispatching to the in-tree HIP/CDNA4 extension
+    (csrc/) — MFMA bf16 GEMMs, fused epilogues.  On a CUDA/HIP device
+    the extension is REQUIRED: there is no silent torch fallback.
+
+Differences from the reference, by design:
+  * reference softmax uses a GLOBAL max and a +1e-7 denominator fudge
+    (functional.py:26); we use the correct per-row max and exact rowsum.
+  * the backward of the loss head is fused: softmax∘MSE and
+    softmax∘cross-entropy each produce d(logits) in one op instead of
+    chaining mse_loss_grad (functional.py:43-44) through softmax_grad
+    (functional.py:30-35).
+"""
+
+import torch
+
+from ._ext import load_ext
+
+
+def _is_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+def _ext_for(t: torch.Tensor):
+    return load_ext(required=True) if _is_gpu(t) else None
+
+
+# ---------------------------------------------------------------- linear
+
+def linear_fwd(x, w, b=None, relu=False):
+    """y = x @ W^T (+ b) (+ ReLU).
+
+    x: (M, K); w: (N, K) row-major ("NT" GEMM — both operands row-major
+    with contiguous K, the natural MFMA layout); b: (N,) or None.
+    Reference: functional.py:13-17 with the module-level fused ReLU of
+    layers.py:120-122 moved into the GEMM epilogue.
+    """
+    if _is_gpu(x):
+        ext = _ext_for(x)
+        empty = torch.Tensor()
+        return ext.gemm_nt(x, w, b if b is not None else empty, empty, bool(relu))
+    y = x @ w.t()
+    if b is not None:
+        y = y + b
+    if relu:
+        y = torch.clamp(y, min=0)
+    return y
+
+
+def linear_dgrad(dy, w, w_t=None, mask_src=None):
+    """dx = (dy ⊙ 1[mask_src>0]) @ W.
+
+    The ReLU backward (reference functional.py:8-10, layers.py:75) is
+    fused into the GEMM A-operand read: when mask_src (the stashed
+    post-ReLU output of THIS layer) is given, dy elements where
+    mask_src<=0 are zeroed during LDS staging.
+
+    GPU path multiplies by the transposed bf16 weight copy w_t ((K, N)
+    row-major) so the kernel is the same NT GEMM as the forward;
+    w_t is maintained by the fused SGD step.
+    Reference: functional.py:20-21 first return value.
+    """
+    if _is_gpu(dy):
+        ext = _ext_for(dy)
+        assert w_t is not None, "GPU dgrad needs the transposed weight copy"
+        empty = torch.Tensor()
+        return ext.gemm_nt(dy, w_t, empty, mask_src if mask_src is not None else empty, False)
+    if mask_src is not None:
+        dy = dy * (mask_src > 0).to(dy.dtype)
+    return dy @ w
+
+
+def linear_wgrad_acc(dy, x, grad_w, grad_b=None, mask_src=None, split_k=0):
+    """grad_w += (dy ⊙ mask)^T @ x ; grad_b += colsum(dy ⊙ mask).
+
+    Accumulates IN PLACE into f32 grad buffers — this is how both
+    µbatch gradient accumulation (reference layers.py:135-136) and
+    split-K parallelism work: the GPU kernel uses f32 atomicAdd, so a
+    K-split over the batch dimension and accumulation across µbatches
+    compose for free.  split_k=0 lets the kernel pick; split_k=1 is the
+    deterministic single-owner path.
+    Reference: functional.py:21 (dW = dout.T @ x, db = dout.sum(0)).
+    """
+    if _is_gpu(dy):
+        ext = _ext_for(dy)
+        ext.wgrad_tn(
+            dy,
+            x,
+            grad_w,
+            grad_b if grad_b is not None else torch.Tensor(),
+            mask_src if mask_src is not None else torch.Tensor(),
+            int(split_k),
+        )
+        return
+    if mask_src is not None:
+        dy = dy * (mask_src > 0).to(dy.dtype)
+    grad_w += (dy.t() @ x).to(grad_w.dtype)
+    if grad_b is not None:
+        grad_b += dy.sum(dim=0).to(grad_b.dtype)
+
+
+# ---------------------------------------------------------------- relu
+
+def relu_fwd(x):
+    """Standalone ReLU (reference functional.py:4-5); the hot path uses
+    the fused GEMM epilogue instead."""
+    if _is_gpu(x):
+        return _ext_for(x).relu_fwd(x)
+    return torch.clamp(x, min=0)
+
+
+def relu_bwd(dy, y):
+    """dx = dy ⊙ 1[y>0] from the stashed ReLU OUTPUT (for relu,
+    out>0 ⟺ in>0, so stashing the output replaces the reference's
+    bitmask stash at layers.py:70)."""
+    if _is_gpu(dy):
+        return _ext_for(dy).relu_bwd(dy, y)
+    return dy * (y > 0).to(dy.dtype)
+
+
+# ---------------------------------------------------------------- softmax
+
+def softmax_fwd(x):
+    """Row softmax with per-row max subtraction.
+
+    Reference: functional.py:24-27 (which uses a global max and +1e-7;
+    we use the numerically correct form — divergence documented)."""
+    if _is_gpu(x):
+        return _ext_for(x).softmax_fwd(x)
+    m = x.max(dim=-1, keepdim=True).values
+    e = torch.exp(x - m)
+    return e / e.sum(dim=-1, keepdim=True)
+
+
+def softmax_bwd(dy, s):
+    """ds/dx given softmax OUTPUT s (we stash the output; the reference
+    stashes the input and recomputes, noted wasteful at
+    functional.py:31-32).  dx = s*(dy - rowsum(s*dy))."""
+    if _is_gpu(dy):
+        return _ext_for(dy).softmax_bwd(dy, s)
+    dot = (s * dy).sum(dim=-1, keepdim=True)
+    return s * (dy - dot)
+
+
+# ---------------------------------------------------------------- loss heads
+
+def mse_loss(x, t, global_batch):
+    """Σ(t−x)²/global_batch (reference functional.py:38-40; only used
+    by tests/logging — training fwd is identity, layers.py:150-155)."""
+    return ((t - x) ** 2).sum() / global_batch
+
+
+def head_softmax_mse_bwd(probs, target, global_batch):
+    """Fused backward of Softmax→MSELoss w.r.t. logits.
+
+    g = −2(t−s)/GB (mse_loss_grad, functional.py:43-44), then the
+    softmax Jacobian (functional.py:30-35):
+      dz = s ⊙ (g − rowsum(s ⊙ g)).
+    global_batch is the GLOBAL batch size so µbatch/DP gradients sum to
+    the sequential gradient (reference layers.py:146-148).
+    """
+    if _is_gpu(probs):
+        return _ext_for(probs).head_mse_bwd(probs, target, float(global_batch))
+    g = -2.0 * (target - probs) / global_batch
+    dot = (probs * g).sum(dim=-1, keepdim=True)
+    return probs * (g - dot)
+
+
+def head_softmax_xent_bwd(probs, target, global_batch):
+    """Fused backward of softmax cross-entropy: dz = (s − t)/GB."""
+    if _is_gpu(probs):
+        return _ext_for(probs).head_xent_bwd(probs, target, float(global_batch))
+    return (probs - target) / global_batch
+
+
+def xent_loss(probs, target, global_batch, eps=1e-9):
+    """−Σ t·log(p)/GB (logging only)."""
+    return -(target * torch.log(probs.float() + eps)).sum() / global_batch

@@ -1,0 +1,11 @@
+from . import instructions  # noqa: F401
+from .comm import GradReducer, Topology, init_topology  # noqa: F401
+from .schedules import (  # noqa: F401
+    SCHEDULES,
+    GPipeSchedule,
+    InferenceSchedule,
+    NaiveParallelSchedule,
+    PipeDreamFlushSchedule,
+    Schedule,
+)
+from .worker import Worker  # noqa: F401

@@ -1,0 +1,218 @@
+"""The Worker: interprets instruction streams against buffers, a model,
+a dataset, an optimizer and the communicator grid.
+
+Reference: shallowspeed/pipe.py:330-466.  Parity notes:
+  * buffer loads with shape asserts (pipe.py:355-365),
+  * PP p2p to stage±1 neighbors (pipe.py:367-381, 414-418),
+  * forward maps input→output buffer (pipe.py:383-387),
+  * backward_and_reduce installs DP hooks around the backward then
+    resets them (pipe.py:389-400),
+  * _INSTRUCTION_MAP dispatch (pipe.py:420-432),
+  * PERSISTENT device buffers — the reference reallocates per batch and
+    carries a TODO to persist them (pipe.py:444-445,446-454); we
+    allocate once per (schedule shape) and reuse.
+
+MI355X notes: buffers are device-resident bf16; p2p tensors go over
+RCCL/xGMI; the DP all-reduce rides torch.distributed's comm stream and
+overlaps backward kernels.
+"""
+
+from typing import Optional
+
+import torch
+
+from ..utils import StepTimer
+from . import comm as comm_mod
+from .comm import GradReducer, Topology
+from .instructions import (
+    BackwardGradAcc,
+    BackwardGradAllReduce,
+    Forward,
+    LoadMuBatchInput,
+    LoadMuBatchTarget,
+    OptimizerStep,
+    RecvActivations,
+    RecvOutputGrad,
+    SendActivations,
+    SendInputGrad,
+    ZeroGrad,
+)
+
+
+class Worker:
+    def __init__(self, topo: Topology, model, dataset=None, optimizer=None,
+                 use_dp: bool = True, bucket_bytes: int = 25 << 20):
+        self.topo = topo
+        self.model = model
+        self.dataset = dataset
+        self.optimizer = optimizer
+        self.device = getattr(model, "device", torch.device("cpu"))
+        self.compute_dtype = getattr(model, "compute_dtype", torch.float32)
+        self.reducer = None
+        if use_dp and topo.dp > 1:
+            self.reducer = GradReducer(model, topo.dp_group, bucket_bytes)
+        self._in_bufs = []
+        self._out_bufs = []
+        self._buf_shape = None
+        # async p2p bookkeeping: ("in"|"out", idx) -> pending isend work
+        self._pending_send = {}
+        self.instruction_times = {}  # class name -> seconds (observability)
+        self._timing = False
+
+        self._DISPATCH = {
+            ZeroGrad: self._zero_grad,
+            OptimizerStep: self._optimizer_step,
+            LoadMuBatchInput: self._load_input,
+            LoadMuBatchTarget: self._load_target,
+            RecvActivations: self._recv_activations,
+            SendActivations: self._send_activations,
+            RecvOutputGrad: self._recv_output_grad,
+            SendInputGrad: self._send_input_grad,
+            Forward: self._forward,
+            BackwardGradAcc: self._backward_acc,
+            BackwardGradAllReduce: self._backward_and_reduce,
+        }
+
+    # ------------------------------------------------------------ buffers
+    def _ensure_buffers(self, num_buffers: int, mubatch_size: int):
+        assert num_buffers % 2 == 0, "num_buffers must be even (pipe.py:446)"
+        n = num_buffers // 2
+        shape = (n, mubatch_size)
+        if self._buf_shape == shape:
+            return
+        self._buf_shape = shape
+        kw = dict(dtype=self.compute_dtype, device=self.device)
+        # four pools: activations in/out, gradients in/out.  Separate
+        # grad pools keep the forward and backward p2p streams from
+        # colliding on buffer reuse in 1F1B steady state (the reference
+        # conflates them, pipe.py:446-454, which only its naive/GPipe
+        # orderings tolerate).
+        self._in_bufs = [
+            torch.zeros(mubatch_size, self.model.in_dim, **kw) for _ in range(n)
+        ]
+        self._out_bufs = [
+            torch.zeros(mubatch_size, self.model.out_dim, **kw) for _ in range(n)
+        ]
+        self._gin_bufs = [
+            torch.zeros(mubatch_size, self.model.out_dim, **kw) for _ in range(n)
+        ]
+        self._gout_bufs = [
+            torch.zeros(mubatch_size, self.model.in_dim, **kw) for _ in range(n)
+        ]
+
+    # --------------------------------------------------------- execution
+    def execute(self, schedule, batch_id: int):
+        self._batch_id = batch_id
+        mub = self.dataset.mubatch_size if self.dataset is not None else \
+            self._buf_shape[1]
+        self._ensure_buffers(schedule.num_buffers, mub)
+        for commands in schedule.steps():
+            for cmd in commands:
+                fn = self._DISPATCH[type(cmd)]
+                if self._timing:
+                    with StepTimer(self.device) as t:
+                        fn(cmd)
+                    key = type(cmd).__name__
+                    self.instruction_times[key] = (
+                        self.instruction_times.get(key, 0.0) + t.elapsed
+                    )
+                else:
+                    fn(cmd)
+        self.flush_sends()
+
+    def enable_instruction_timing(self, on: bool = True):
+        self._timing = on
+
+    # ----------------------------------------------------- async p2p
+    # Sends are nonblocking isends (RCCL p2p over the direct xGMI link
+    # between stage-adjacent GPUs); a buffer is only waited on when it
+    # is about to be OVERWRITTEN.  This is the double-buffered async
+    # p2p the reference leaves as a TODO (pipe.py:269-272) and what
+    # makes the 1F1B steady state deadlock-free with blocking recvs.
+    def _wait_buffer(self, kind, idx):
+        h = self._pending_send.pop((kind, idx), None)
+        if h is not None:
+            h.wait()
+
+    def _isend(self, kind, idx, tensor, dst):
+        self._wait_buffer(kind, idx)
+        self._pending_send[(kind, idx)] = comm_mod.isend_tensor(tensor, dst)
+
+    def flush_sends(self):
+        for h in self._pending_send.values():
+            h.wait()
+        self._pending_send.clear()
+
+    # ------------------------------------------------- instruction impls
+    def _zero_grad(self, cmd):
+        self.model.zero_grad()
+
+    def _optimizer_step(self, cmd):
+        self.optimizer.step()
+
+    def _load_input(self, cmd):
+        x = self.dataset.micro_batch_input(self._batch_id, cmd.mubatch_id)
+        self._wait_buffer("in", cmd.buffer_idx)
+        buf = self._in_bufs[cmd.buffer_idx]
+        assert x.shape == buf.shape, (x.shape, buf.shape)  # pipe.py:357-360
+        buf.copy_(x.to(self.compute_dtype), non_blocking=True)
+
+    def _load_target(self, cmd):
+        y = self.dataset.micro_batch_target(self._batch_id, cmd.mubatch_id)
+        self._wait_buffer("gin", cmd.buffer_idx)
+        buf = self._gin_bufs[cmd.buffer_idx]
+        assert y.shape == buf.shape, (y.shape, buf.shape)  # pipe.py:362-365
+        buf.copy_(y.to(self.compute_dtype), non_blocking=True)
+
+    def _recv_activations(self, cmd):
+        self._wait_buffer("in", cmd.buffer_idx)
+        comm_mod.recv_tensor(self._in_bufs[cmd.buffer_idx], self.topo.prev_rank)
+
+    def _send_activations(self, cmd):
+        self._isend("out", cmd.buffer_idx,
+                    self._out_bufs[cmd.buffer_idx], self.topo.next_rank)
+
+    def _recv_output_grad(self, cmd):
+        self._wait_buffer("gin", cmd.buffer_idx)
+        comm_mod.recv_tensor(self._gin_bufs[cmd.buffer_idx], self.topo.next_rank)
+
+    def _send_input_grad(self, cmd):
+        self._isend("gout", cmd.buffer_idx,
+                    self._gout_bufs[cmd.buffer_idx], self.topo.prev_rank)
+
+    def _forward(self, cmd):
+        x = self._in_bufs[cmd.in_buffer]
+        if self.model._training:
+            # Layers stash their input per µbatch for wgrad; the input
+            # buffer is SHARED across µbatches (overwritten by the next
+            # Load/Recv), so snapshot it.  (The reference stashes the
+            # live buffer reference — layers.py:117 + pipe.py:447-454 —
+            # which is only safe for its naive schedule ordering.)
+            x = x.clone()
+        y = self.model.forward(x, cmd.mubatch_id)
+        self._wait_buffer("out", cmd.out_buffer)
+        self._out_bufs[cmd.out_buffer].copy_(y)
+
+    def _backward_acc(self, cmd):
+        d = self.model.backward(self._gin_bufs[cmd.out_buffer], cmd.mubatch_id)
+        # stage 0 never sends input grads, so skip the staging copy
+        if d is not None and self.topo.stage_id != 0:
+            self._wait_buffer("gout", cmd.in_buffer)
+            self._gout_bufs[cmd.in_buffer].copy_(d)
+        return d
+
+    def _backward_and_reduce(self, cmd):
+        """Install DP hooks, run backward, reset hooks — the one
+        'inversion' in the stack (pipe.py:389-400 ↔ layers.py:201-213):
+        grad-ready → bucket all-reduce launches mid-backward."""
+        if self.reducer is not None:
+            self.reducer.reset()
+            self.model.register_grad_hook(self.reducer.param_done)
+            self.model.register_post_grad_hook(
+                lambda params: self.reducer.finalize())
+        try:
+            self._backward_acc(cmd)
+        finally:
+            if self.reducer is not None:
+                self.model.reset_grad_hooks()
+                self.model.reset_post_grad_hooks()

@@ -1,0 +1,148 @@
+"""CLI training driver.
+
+Reference: /root/reference/train.py — identical core CLI
+(`--dp N --pp M --schedule {naive,gpipe,pipedream}`, train.py:62-74)
+plus flags for batch/µbatch/layer/loss/dtype/synthetic-data knobs
+(the reference hardwires EPOCHS=20, GLOBAL_BATCH_SIZE=128,
+N_MUBATCHES=4, lr=0.006, layer_sizes=[784,128,...,10] at
+train.py:56-59,98,107).
+
+Launch (multi-GPU, one rank per GPU over RCCL):
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 train.py --dp D --pp P ...   (D*P == N)
+
+Single process: python train.py          (dp=1, pp=1, CPU or 1 GPU)
+"""
+
+import argparse
+import time
+
+import torch
+
+from shallowspeed_amd.data import Dataset
+from shallowspeed_amd.models import MLP, SGD
+from shallowspeed_amd.parallel import SCHEDULES, InferenceSchedule, Worker, init_topology
+from shallowspeed_amd.utils import assert_sync, get_model_hash, rprint
+
+
+def compute_accuracy(model, worker, dataset, topo):
+    """Forward-only eval over the val split; returns accuracy on the
+    last stage, None elsewhere.  Reference: train.py:21-47 (argmax of
+    the output buffer vs argmax of target, train.py:40-43)."""
+    model.eval()
+    correct, total = 0, 0
+    sched = InferenceSchedule(1, topo.pp, topo.stage_id)
+    for b in range(dataset.num_batches()):
+        worker.execute(sched, b)
+        if topo.stage_id == topo.pp - 1:
+            probs = worker._out_bufs[0]
+            target = dataset.micro_batch_target(b, 0)
+            pred = probs.float().argmax(dim=-1)
+            lab = target.to(probs.device).float().argmax(dim=-1)
+            correct += (pred == lab).sum().item()
+            total += pred.numel()
+    model.train()
+    if topo.stage_id != topo.pp - 1:
+        return None
+    if topo.dp > 1:
+        t = torch.tensor([correct, total], dtype=torch.float64)
+        import torch.distributed as dist
+
+        if topo.device.type == "cuda":
+            t = t.to(topo.device)
+        dist.all_reduce(t, group=topo.dp_group)
+        correct, total = t[0].item(), t[1].item()
+    return correct / max(total, 1)
+
+
+def parse_sizes(s):
+    return [int(v) for v in s.split(",")]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--dp", type=int, default=1)
+    ap.add_argument("--pp", type=int, default=1)
+    ap.add_argument("--schedule", choices=list(SCHEDULES), default="gpipe")
+    ap.add_argument("--epochs", type=int, default=20)
+    ap.add_argument("--global-batch", type=int, default=128)
+    ap.add_argument("--mubatches", type=int, default=4)
+    ap.add_argument("--lr", type=float, default=0.006)
+    ap.add_argument("--layer-sizes", type=parse_sizes,
+                    default=[784, 128, 127, 126, 125, 124, 123, 10],
+                    help="comma-separated boundaries (reference train.py:98)")
+    ap.add_argument("--loss", choices=["xent", "mse"], default="mse",
+                    help="loss head; mse matches the reference head")
+    ap.add_argument("--data-dir", default=None,
+                    help="load x_*.parquet / y_*.npy like the reference "
+                         "if present; otherwise synthetic")
+    ap.add_argument("--samples", type=int, default=8192)
+    ap.add_argument("--device", default=None, help="cpu | cuda")
+    ap.add_argument("--save", default=None, help="checkpoint dir to write")
+    ap.add_argument("--resume", default=None, help="checkpoint dir to read")
+    args = ap.parse_args()
+
+    device = torch.device(args.device) if args.device else None
+    topo = init_topology(args.dp, args.pp, device=device)
+    device = topo.device
+
+    # model: stage slice of the full MLP (reference train.py:99-107)
+    model = MLP(args.layer_sizes, stage_idx=topo.stage_id, n_stages=args.pp,
+                global_batch_size=args.global_batch, loss=args.loss)
+    model.materialize_device(device)
+    optimizer = SGD(model.parameters(), lr=args.lr)
+
+    if args.resume:
+        from shallowspeed_amd.checkpoint import load_checkpoint
+
+        load_checkpoint(args.resume, model, topo)
+
+    mubatch = args.global_batch // args.dp // args.mubatches
+    train_ds = Dataset(args.global_batch, mubatch, save_dir=args.data_dir,
+                       n_samples=args.samples, in_dim=args.layer_sizes[0],
+                       n_classes=args.layer_sizes[-1], device=device)
+    train_ds.load(topo.dp_rank, args.dp)
+    val_batch = min(args.samples // 4, 1024)
+    val_ds = Dataset(val_batch * args.dp, val_batch, save_dir=args.data_dir,
+                     validation=True, n_samples=max(args.samples // 4, val_batch * args.dp),
+                     in_dim=args.layer_sizes[0],
+                     n_classes=args.layer_sizes[-1], device=device)
+    val_ds.load(topo.dp_rank, args.dp)
+
+    worker = Worker(topo, model, train_ds, optimizer)
+    val_worker = Worker(topo, model, val_ds, None, use_dp=False)
+    sched_cls = SCHEDULES[args.schedule]
+
+    rprint(f"world={topo.world} dp={args.dp} pp={args.pp} "
+           f"schedule={args.schedule} device={device} "
+           f"layers={args.layer_sizes} loss={args.loss}")
+
+    for epoch in range(args.epochs):
+        t0 = time.time()
+        acc = compute_accuracy(model, val_worker, val_ds, topo)
+        for batch_id in range(train_ds.num_batches()):
+            sched = sched_cls(train_ds.num_mubatches(), args.pp, topo.stage_id)
+            worker.execute(sched, batch_id)
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+        if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0:
+            print(f"epoch {epoch:3d}  val_acc={acc:.4f}  "
+                  f"time={time.time()-t0:.2f}s", flush=True)
+
+    acc = compute_accuracy(model, val_worker, val_ds, topo)
+    if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0:
+        print(f"final val_acc={acc:.4f}", flush=True)
+
+    if args.save:
+        from shallowspeed_amd.checkpoint import save_checkpoint
+
+        save_checkpoint(args.save, model, topo, step=args.epochs)
+
+    # replica-sync invariant (reference train.py:154-155)
+    if topo.dp > 1:
+        assert_sync(topo.dp_group, get_model_hash(model))
+        rprint("DP replicas in sync ✓")
+
+
+if __name__ == "__main__":
+    main()

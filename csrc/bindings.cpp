@@ -1,0 +1,177 @@
+// Python bindings for the shallowspeed_amd HIP/CDNA4 kernels.
+//
+// Thin torch-extension layer: shape/dtype/contiguity checks + launch on
+// the current torch stream.  All compute lives in gemm.hip /
+// elementwise.hip (raw HIP, extern "C" launchers).
+
+#include <torch/extension.h>
+
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void ss_gemm_nt(const void*, const void*, const void*, const void*, void*,
+                int, int, int, bool, hipStream_t);
+void ss_wgrad_tn(const void*, const void*, const void*, void*, int, int, int,
+                 int, hipStream_t);
+void ss_colsum(const void*, const void*, void*, int, int, hipStream_t);
+void ss_relu_fwd(const void*, void*, long, hipStream_t);
+void ss_relu_bwd(const void*, const void*, void*, long, hipStream_t);
+void ss_softmax_fwd(const void*, void*, int, int, hipStream_t);
+void ss_softmax_bwd(const void*, const void*, void*, int, int, hipStream_t);
+void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
+                     hipStream_t);
+void ss_head_xent_bwd(const void*, const void*, void*, long, float,
+                      hipStream_t);
+void ss_sgd_multi(const void*, int, long, float, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+    return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == torch::kFloat, name, " must be f32");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+bool has(const torch::Tensor& t) { return t.defined() && t.numel() > 0; }
+
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor bias,
+                      torch::Tensor mask, bool relu) {
+    check_bf16(a, "a");
+    check_bf16(b, "b");
+    TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "a/b must be 2-D");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "K mismatch: ", K, " vs ", b.size(1));
+    const void* bias_p = nullptr;
+    if (has(bias)) {
+        check_bf16(bias, "bias");
+        TORCH_CHECK(bias.numel() == N, "bias size");
+        bias_p = bias.data_ptr();
+    }
+    const void* mask_p = nullptr;
+    if (has(mask)) {
+        check_bf16(mask, "mask");
+        TORCH_CHECK(mask.sizes() == a.sizes(), "mask must match A");
+        mask_p = mask.data_ptr();
+    }
+    auto c = torch::empty({M, N}, a.options());
+    ss_gemm_nt(a.data_ptr(), b.data_ptr(), bias_p, mask_p, c.data_ptr(), M, N,
+               K, relu, cur_stream());
+    return c;
+}
+
+void wgrad_tn(torch::Tensor dy, torch::Tensor x, torch::Tensor gw,
+              torch::Tensor gb, torch::Tensor mask, int64_t split_k) {
+    check_bf16(dy, "dy");
+    check_bf16(x, "x");
+    check_f32(gw, "gw");
+    const int Kb = dy.size(0), Mo = dy.size(1), N = x.size(1);
+    TORCH_CHECK(x.size(0) == Kb, "batch mismatch");
+    TORCH_CHECK(gw.size(0) == Mo && gw.size(1) == N, "gw shape");
+    const void* mask_p = nullptr;
+    if (has(mask)) {
+        check_bf16(mask, "mask");
+        TORCH_CHECK(mask.sizes() == dy.sizes(), "mask must match dy");
+        mask_p = mask.data_ptr();
+    }
+    ss_wgrad_tn(dy.data_ptr(), x.data_ptr(), mask_p, gw.data_ptr(), Mo, N, Kb,
+                (int)split_k, cur_stream());
+    if (has(gb)) {
+        check_f32(gb, "gb");
+        TORCH_CHECK(gb.numel() == Mo, "gb size");
+        ss_colsum(dy.data_ptr(), mask_p, gb.data_ptr(), Kb, Mo, cur_stream());
+    }
+}
+
+torch::Tensor relu_fwd(torch::Tensor x) {
+    check_bf16(x, "x");
+    auto y = torch::empty_like(x);
+    ss_relu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), cur_stream());
+    return y;
+}
+
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
+    check_bf16(dy, "dy");
+    check_bf16(y, "y");
+    TORCH_CHECK(dy.sizes() == y.sizes(), "shape mismatch");
+    auto dx = torch::empty_like(dy);
+    ss_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(), dy.numel(),
+                cur_stream());
+    return dx;
+}
+
+torch::Tensor softmax_fwd(torch::Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 2, "x must be 2-D");
+    auto s = torch::empty_like(x);
+    ss_softmax_fwd(x.data_ptr(), s.data_ptr(), x.size(0), x.size(1),
+                   cur_stream());
+    return s;
+}
+
+torch::Tensor softmax_bwd(torch::Tensor dy, torch::Tensor s) {
+    check_bf16(dy, "dy");
+    check_bf16(s, "s");
+    TORCH_CHECK(dy.sizes() == s.sizes(), "shape mismatch");
+    auto dx = torch::empty_like(dy);
+    ss_softmax_bwd(dy.data_ptr(), s.data_ptr(), dx.data_ptr(), dy.size(0),
+                   dy.size(1), cur_stream());
+    return dx;
+}
+
+torch::Tensor head_mse_bwd(torch::Tensor s, torch::Tensor t,
+                           double global_batch) {
+    check_bf16(s, "probs");
+    check_bf16(t, "target");
+    TORCH_CHECK(s.sizes() == t.sizes(), "shape mismatch");
+    auto dz = torch::empty_like(s);
+    ss_head_mse_bwd(s.data_ptr(), t.data_ptr(), dz.data_ptr(), s.size(0),
+                    s.size(1), (float)(1.0 / global_batch), cur_stream());
+    return dz;
+}
+
+torch::Tensor head_xent_bwd(torch::Tensor s, torch::Tensor t,
+                            double global_batch) {
+    check_bf16(s, "probs");
+    check_bf16(t, "target");
+    TORCH_CHECK(s.sizes() == t.sizes(), "shape mismatch");
+    auto dz = torch::empty_like(s);
+    ss_head_xent_bwd(s.data_ptr(), t.data_ptr(), dz.data_ptr(), s.numel(),
+                     (float)(1.0 / global_batch), cur_stream());
+    return dz;
+}
+
+void sgd_multi(torch::Tensor desc, double lr, int64_t total) {
+    // total passed by the caller (cached host-side) — no device sync.
+    TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
+                    desc.is_contiguous() && desc.dim() == 2 &&
+                    desc.size(1) == 7,
+                "desc must be CUDA int64 [T,7]");
+    ss_sgd_multi(desc.data_ptr(), (int)desc.size(0), (long)total, (float)lr,
+                 cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
+    m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum");
+    m.def("relu_fwd", &relu_fwd);
+    m.def("relu_bwd", &relu_bwd);
+    m.def("softmax_fwd", &softmax_fwd);
+    m.def("softmax_bwd", &softmax_bwd);
+    m.def("head_mse_bwd", &head_mse_bwd);
+    m.def("head_xent_bwd", &head_xent_bwd);
+    m.def("sgd_multi", &sgd_multi);
+}

@@ -1,0 +1,249 @@
+// Elementwise / rowwise / optimizer kernels (gfx950, MI355X).
+//
+// Replaces the reference's NumPy ops:
+//   * relu fwd/bwd            (functional.py:4-10; hot path uses the
+//                              fused GEMM epilogue/mask instead)
+//   * row softmax fwd/bwd     (functional.py:24-35 — with CORRECT
+//                              per-row max, unlike the reference's
+//                              global-max quirk at functional.py:26)
+//   * fused loss heads        (softmax∘MSE per functional.py:43-44 +
+//                              softmax jacobian; softmax-cross-entropy)
+//   * multi-tensor SGD        (optimizer.py:10-13) — ONE launch updates
+//                              every stage parameter: f32 master -=
+//                              lr·f32 grad, re-emitting the bf16 copy
+//                              AND the transposed bf16 copy (which
+//                              makes dgrad an NT GEMM).
+//
+// All memory-bound: bf16 traffic vectorized as ushort8 (16 B/lane,
+// Guideline 13), f32 math internally, wave-shuffle row reductions.
+
+#include "common.h"
+
+// ------------------------------------------------------------- relu
+
+__global__ __launch_bounds__(256) void relu_fwd_kernel(
+    const __bf16* __restrict__ x, __bf16* __restrict__ y, long n) {
+    long i = (long)(blockIdx.x * 256 + threadIdx.x) * 8;
+    const long stride = (long)gridDim.x * 256 * 8;
+    for (; i + 8 <= n; i += stride) {
+        bf16x8 v = *(const bf16x8*)(x + i);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = bf2f(v[j]);
+            o[j] = f2bf(f > 0.f ? f : 0.f);
+        }
+        *(bf16x8*)(y + i) = o;
+    }
+    if (i < n) {
+        for (; i < n; ++i) {
+            float f = bf2f(x[i]);
+            y[i] = f2bf(f > 0.f ? f : 0.f);
+        }
+    }
+}
+
+__global__ __launch_bounds__(256) void relu_bwd_kernel(
+    const __bf16* __restrict__ dy, const __bf16* __restrict__ y,
+    __bf16* __restrict__ dx, long n) {
+    long i = (long)(blockIdx.x * 256 + threadIdx.x) * 8;
+    const long stride = (long)gridDim.x * 256 * 8;
+    for (; i + 8 <= n; i += stride) {
+        bf16x8 g = *(const bf16x8*)(dy + i);
+        bf16x8 m = *(const bf16x8*)(y + i);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            o[j] = bf2f(m[j]) > 0.f ? g[j] : (__bf16)0.f;
+        *(bf16x8*)(dx + i) = o;
+    }
+    if (i < n) {
+        for (; i < n; ++i)
+            dx[i] = bf2f(y[i]) > 0.f ? dy[i] : (__bf16)0.f;
+    }
+}
+
+// ------------------------------------------------------- row reductions
+
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        v = fmaxf(v, __shfl_xor(v, off, 64));
+    return v;
+}
+
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        v += __shfl_xor(v, off, 64);
+    return v;
+}
+
+// one wave per row; 4 rows per block
+__global__ __launch_bounds__(256) void softmax_fwd_kernel(
+    const __bf16* __restrict__ x, __bf16* __restrict__ s, int B, int C) {
+    const int lane = threadIdx.x & 63;
+    const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= B) return;
+    const __bf16* xr = x + (long)row * C;
+    float m = -1e30f;
+    for (int c = lane; c < C; c += 64) m = fmaxf(m, bf2f(xr[c]));
+    m = wave_max(m);
+    float sum = 0.f;
+    for (int c = lane; c < C; c += 64) sum += __expf(bf2f(xr[c]) - m);
+    sum = wave_sum(sum);
+    const float inv = 1.f / sum;
+    __bf16* sr = s + (long)row * C;
+    for (int c = lane; c < C; c += 64)
+        sr[c] = f2bf(__expf(bf2f(xr[c]) - m) * inv);
+}
+
+// dx = s * (dy - rowsum(s*dy)), from stashed OUTPUT s
+__global__ __launch_bounds__(256) void softmax_bwd_kernel(
+    const __bf16* __restrict__ dy, const __bf16* __restrict__ s,
+    __bf16* __restrict__ dx, int B, int C) {
+    const int lane = threadIdx.x & 63;
+    const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= B) return;
+    const __bf16* dyr = dy + (long)row * C;
+    const __bf16* sr = s + (long)row * C;
+    float dot = 0.f;
+    for (int c = lane; c < C; c += 64) dot += bf2f(sr[c]) * bf2f(dyr[c]);
+    dot = wave_sum(dot);
+    __bf16* dxr = dx + (long)row * C;
+    for (int c = lane; c < C; c += 64)
+        dxr[c] = f2bf(bf2f(sr[c]) * (bf2f(dyr[c]) - dot));
+}
+
+// fused softmax∘MSE head backward: g = -2(t-s)/GB;
+// dz = s * (g - rowsum(s*g))   (reference functional.py:30-35,43-44)
+__global__ __launch_bounds__(256) void head_mse_bwd_kernel(
+    const __bf16* __restrict__ s, const __bf16* __restrict__ t,
+    __bf16* __restrict__ dz, int B, int C, float inv_gb) {
+    const int lane = threadIdx.x & 63;
+    const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= B) return;
+    const __bf16* sr = s + (long)row * C;
+    const __bf16* tr = t + (long)row * C;
+    float dot = 0.f;
+    for (int c = lane; c < C; c += 64) {
+        const float sv = bf2f(sr[c]);
+        const float g = -2.f * (bf2f(tr[c]) - sv) * inv_gb;
+        dot += sv * g;
+    }
+    dot = wave_sum(dot);
+    __bf16* dzr = dz + (long)row * C;
+    for (int c = lane; c < C; c += 64) {
+        const float sv = bf2f(sr[c]);
+        const float g = -2.f * (bf2f(tr[c]) - sv) * inv_gb;
+        dzr[c] = f2bf(sv * (g - dot));
+    }
+}
+
+// fused softmax-cross-entropy backward: dz = (s - t)/GB
+__global__ __launch_bounds__(256) void head_xent_bwd_kernel(
+    const __bf16* __restrict__ s, const __bf16* __restrict__ t,
+    __bf16* __restrict__ dz, long n, float inv_gb) {
+    long i = (long)(blockIdx.x * 256 + threadIdx.x) * 8;
+    const long stride = (long)gridDim.x * 256 * 8;
+    for (; i + 8 <= n; i += stride) {
+        bf16x8 sv = *(const bf16x8*)(s + i);
+        bf16x8 tv = *(const bf16x8*)(t + i);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            o[j] = f2bf((bf2f(sv[j]) - bf2f(tv[j])) * inv_gb);
+        *(bf16x8*)(dz + i) = o;
+    }
+    if (i < n)
+        for (; i < n; ++i)
+            dz[i] = f2bf((bf2f(s[i]) - bf2f(t[i])) * inv_gb);
+}
+
+// --------------------------------------------------------- fused SGD
+
+// desc row (int64 x 7): master*, grad*, lp*, lpt*, numel, cols, start
+__global__ __launch_bounds__(256) void sgd_multi_kernel(
+    const long* __restrict__ desc, int ntensors, long total, float lr) {
+    long e = (long)blockIdx.x * 256 + threadIdx.x;
+    const long stride = (long)gridDim.x * 256;
+    for (; e < total; e += stride) {
+        // binary search the tensor whose [start, start+numel) holds e
+        int lo = 0, hi = ntensors - 1;
+        while (lo < hi) {
+            const int mid = (lo + hi + 1) >> 1;
+            if (desc[mid * 7 + 6] <= e) lo = mid;
+            else hi = mid - 1;
+        }
+        const long* d = desc + lo * 7;
+        const long i = e - d[6];
+        float* master = (float*)d[0];
+        const float* grad = (const float*)d[1];
+        const float m = master[i] - lr * grad[i];
+        master[i] = m;
+        const __bf16 v = f2bf(m);
+        if (d[2]) ((__bf16*)d[2])[i] = v;
+        if (d[3]) {
+            const long cols = d[5];
+            const long rows = d[4] / cols;
+            ((__bf16*)d[3])[(i % cols) * rows + i / cols] = v;
+        }
+    }
+}
+
+// ---------------------------------------------------------------- launchers
+
+extern "C" {
+
+static inline int ew_grid(long n) {
+    long blocks = (n / 8 + 255) / 256;
+    if (blocks < 1) blocks = 1;
+    if (blocks > 2048) blocks = 2048;  // grid-stride the rest (G11)
+    return (int)blocks;
+}
+
+void ss_relu_fwd(const void* x, void* y, long n, hipStream_t st) {
+    hipLaunchKernelGGL(relu_fwd_kernel, dim3(ew_grid(n)), dim3(256), 0, st,
+                       (const __bf16*)x, (__bf16*)y, n);
+}
+
+void ss_relu_bwd(const void* dy, const void* y, void* dx, long n,
+                 hipStream_t st) {
+    hipLaunchKernelGGL(relu_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0, st,
+                       (const __bf16*)dy, (const __bf16*)y, (__bf16*)dx, n);
+}
+
+void ss_softmax_fwd(const void* x, void* s, int B, int C, hipStream_t st) {
+    hipLaunchKernelGGL(softmax_fwd_kernel, dim3(cdiv(B, 4)), dim3(256), 0, st,
+                       (const __bf16*)x, (__bf16*)s, B, C);
+}
+
+void ss_softmax_bwd(const void* dy, const void* s, void* dx, int B, int C,
+                    hipStream_t st) {
+    hipLaunchKernelGGL(softmax_bwd_kernel, dim3(cdiv(B, 4)), dim3(256), 0, st,
+                       (const __bf16*)dy, (const __bf16*)s, (__bf16*)dx, B, C);
+}
+
+void ss_head_mse_bwd(const void* s, const void* t, void* dz, int B, int C,
+                     float inv_gb, hipStream_t st) {
+    hipLaunchKernelGGL(head_mse_bwd_kernel, dim3(cdiv(B, 4)), dim3(256), 0, st,
+                       (const __bf16*)s, (const __bf16*)t, (__bf16*)dz, B, C,
+                       inv_gb);
+}
+
+void ss_head_xent_bwd(const void* s, const void* t, void* dz, long n,
+                      float inv_gb, hipStream_t st) {
+    hipLaunchKernelGGL(head_xent_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                       st, (const __bf16*)s, (const __bf16*)t, (__bf16*)dz, n,
+                       inv_gb);
+}
+
+void ss_sgd_multi(const void* desc, int ntensors, long total, float lr,
+                  hipStream_t st) {
+    long blocks = (total + 255) / 256;
+    if (blocks > 1024) blocks = 1024;
+    hipLaunchKernelGGL(sgd_multi_kernel, dim3((int)blocks), dim3(256), 0, st,
+                       (const long*)desc, ntensors, total, lr);
+}
+
+}  // extern "C"

@@ -1,0 +1,376 @@
+// MFMA bf16 GEMM kernels for the MLP hot path (gfx950 / MI355X).
+//
+// Replaces the reference's NumPy BLAS calls (functional.py:13-21):
+//   * gemm_nt:  C[M,N] = A[M,K] · B[N,K]^T (+bias) (+ReLU epilogue)
+//               — forward (A=x, B=W) AND dgrad (A=dy, B=W^T copy),
+//               with an optional fused ReLU-backward mask applied to A
+//               while staging (mask = stashed post-ReLU output, same
+//               shape as A).
+//   * wgrad_tn: gW[Mo,N] += (A[Kb,Mo] ⊙ mask)^T · B[Kb,N], split-K over
+//               the batch with f32 atomicAdd — the atomic IS the
+//               gradient accumulation across µbatches/split-K slices
+//               (reference grad accumulation: layers.py:135-136).
+//   * colsum:   gb[N] += Σ_rows (dY ⊙ mask) — bias grad.
+//
+// Design notes (cdna_hip_programming.md):
+//   * mfma_f32_16x16x32_bf16: per-wave 16×16 tile, K=32 per issue;
+//     lane l holds A[l&15][(l>>4)*8 + j] / B-col[l&15], C row
+//     (l>>4)*4+reg, col l&15 (verified on hardware by
+//     tests/test_gpu_numerics.py::test_mfma_layout).
+//   * LDS tiles padded +8 bf16 per row (16 B) → conflict-free
+//     ds_read_b128 fragment reads (Guideline 4).
+//   * These MLP shapes are small/memory-bound; the simple one-buffer
+//     two-barrier structure is chosen for robustness at arbitrary
+//     M/N/K (full bounds handling) — launch count, not MFMA peak, is
+//     the budget here.
+
+#include "common.h"
+
+#define BK 32
+#define LDS_PAD 8
+
+// ---------------------------------------------------------------- gemm_nt
+
+template <int BM, int BN, int WAVES_M, int WAVES_N,
+          bool HAS_BIAS, bool RELU, bool HAS_MASK>
+__global__ __launch_bounds__(256) void gemm_nt_kernel(
+    const __bf16* __restrict__ A,     // [M][K]
+    const __bf16* __restrict__ B,     // [N][K]
+    const __bf16* __restrict__ bias,  // [N]
+    const __bf16* __restrict__ mask,  // [M][K] (A-mask source, >0 keeps)
+    __bf16* __restrict__ C,           // [M][N]
+    int M, int N, int K) {
+    constexpr int WM = BM / WAVES_M;        // wave tile rows
+    constexpr int WN = BN / WAVES_N;        // wave tile cols
+    constexpr int FM = WM / 16;             // 16x16 frags per wave (M)
+    constexpr int FN = WN / 16;
+    constexpr int LDA = BK + LDS_PAD;
+
+    __shared__ ushort As[BM][LDA];
+    __shared__ ushort Bs[BN][LDA];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = wave / WAVES_N;
+    const int wn = wave % WAVES_N;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kch = lane >> 4;  // 0..3 -> k offset kch*8
+
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        // ---- stage A tile [BM][BK] (row-major slice of A) ----
+        {
+            constexpr int EL = BM * BK / 256;  // elems per thread (4 or 8)
+            const int off = tid * EL;
+            const int r = off / BK, c = off % BK;
+            const int gm = m0 + r, gk = k0 + c;
+            if (gm < M && gk + EL <= K && !HAS_MASK) {
+                // fast path: contiguous vector load/store
+                if constexpr (EL == 8) {
+                    *(ulonglong2*)&As[r][c] =
+                        *(const ulonglong2*)&A[(long)gm * K + gk];
+                } else {
+                    *(unsigned long long*)&As[r][c] =
+                        *(const unsigned long long*)&A[(long)gm * K + gk];
+                }
+            } else {
+#pragma unroll
+                for (int i = 0; i < EL; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    if (gm < M && gk + i < K) {
+                        v = A[(long)gm * K + gk + i];
+                        if constexpr (HAS_MASK) {
+                            if (!(bf2f(mask[(long)gm * K + gk + i]) > 0.f))
+                                v = (__bf16)0.f;
+                        }
+                    }
+                    As[r][c + i] = *(const ushort*)&v;
+                }
+            }
+            if constexpr (HAS_MASK) {
+                // mask path above is scalar; also apply on fast path
+                // (handled by taking the scalar branch when HAS_MASK)
+            }
+        }
+        // ---- stage B tile [BN][BK] ----
+        {
+            constexpr int EL = BN * BK / 256;
+            const int off = tid * EL;
+            const int r = off / BK, c = off % BK;
+            const int gn = n0 + r, gk = k0 + c;
+            if (gn < N && gk + EL <= K) {
+                if constexpr (EL == 8) {
+                    *(ulonglong2*)&Bs[r][c] =
+                        *(const ulonglong2*)&B[(long)gn * K + gk];
+                } else {
+                    *(unsigned long long*)&Bs[r][c] =
+                        *(const unsigned long long*)&B[(long)gn * K + gk];
+                }
+            } else {
+#pragma unroll
+                for (int i = 0; i < EL; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    if (gn < N && gk + i < K) v = B[(long)gn * K + gk + i];
+                    Bs[r][c + i] = *(const ushort*)&v;
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- MFMA ----
+        bf16x8 a_frag[FM], b_frag[FN];
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+            a_frag[i] = *(const bf16x8*)&As[wm * WM + i * 16 + lrow][kch * 8];
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+            b_frag[j] = *(const bf16x8*)&Bs[wn * WN + j * 16 + lrow][kch * 8];
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int j = 0; j < FN; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+    // ---- epilogue: bias, ReLU, bf16 store ----
+#pragma unroll
+    for (int i = 0; i < FM; ++i) {
+#pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const int gcol = n0 + wn * WN + j * 16 + lrow;
+            if (gcol >= N) continue;
+            float bv = 0.f;
+            if constexpr (HAS_BIAS) bv = bf2f(bias[gcol]);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int grow = m0 + wm * WM + i * 16 + kch * 4 + r;
+                if (grow >= M) continue;
+                float v = acc[i][j][r];
+                if constexpr (HAS_BIAS) v += bv;
+                if constexpr (RELU) v = v > 0.f ? v : 0.f;
+                C[(long)grow * N + gcol] = f2bf(v);
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------- wgrad
+
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256) void wgrad_tn_kernel(
+    const __bf16* __restrict__ dY,    // [Kb][Mo]
+    const __bf16* __restrict__ X,     // [Kb][N]
+    const __bf16* __restrict__ mask,  // [Kb][Mo]
+    float* __restrict__ gW,           // [Mo][N] (atomicAdd +=)
+    int Mo, int N, int Kb, int k_per_split) {
+    constexpr int BM = 64, BN = 64;
+    constexpr int LDA = BK + LDS_PAD;
+
+    __shared__ ushort At[BM][LDA];  // dY^T slice: [m][k]
+    __shared__ ushort Bt[BN][LDA];  // X^T slice:  [n][k]
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = wave >> 1;  // 2x2 waves
+    const int wn = wave & 1;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int kbeg = blockIdx.z * k_per_split;
+    const int kend = min(Kb, kbeg + k_per_split);
+
+    f32x4 acc[2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kch = lane >> 4;
+
+    for (int k0 = kbeg; k0 < kend; k0 += BK) {
+        // stage dY rows k0..k0+32 transposed into At[m][k]
+        {
+            constexpr int EL = 8;  // 32*64/256
+            const int off = tid * EL;
+            const int kr = off / BM, mc = off % BM;
+            const int gk = k0 + kr;
+#pragma unroll
+            for (int i = 0; i < EL; ++i) {
+                __bf16 v = (__bf16)0.f;
+                const int gm = m0 + mc + i;
+                if (gk < kend && gm < Mo) {
+                    v = dY[(long)gk * Mo + gm];
+                    if constexpr (HAS_MASK) {
+                        if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
+                            v = (__bf16)0.f;
+                    }
+                }
+                At[mc + i][kr] = *(const ushort*)&v;
+            }
+        }
+        // stage X rows transposed into Bt[n][k]
+        {
+            constexpr int EL = 8;
+            const int off = tid * EL;
+            const int kr = off / BN, nc = off % BN;
+            const int gk = k0 + kr;
+#pragma unroll
+            for (int i = 0; i < EL; ++i) {
+                __bf16 v = (__bf16)0.f;
+                const int gn = n0 + nc + i;
+                if (gk < kend && gn < N) v = X[(long)gk * N + gn];
+                Bt[nc + i][kr] = *(const ushort*)&v;
+            }
+        }
+        __syncthreads();
+
+        bf16x8 a_frag[2], b_frag[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+            a_frag[i] = *(const bf16x8*)&At[wm * 32 + i * 16 + lrow][kch * 8];
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+            b_frag[j] = *(const bf16x8*)&Bt[wn * 32 + j * 16 + lrow][kch * 8];
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            const int gcol = n0 + wn * 32 + j * 16 + lrow;
+            if (gcol >= N) continue;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int grow = m0 + wm * 32 + i * 16 + kch * 4 + r;
+                if (grow >= Mo) continue;
+                atomicAdd(&gW[(long)grow * N + gcol], acc[i][j][r]);
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------- colsum
+
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256) void colsum_kernel(
+    const __bf16* __restrict__ dY,    // [M][N]
+    const __bf16* __restrict__ mask,  // [M][N]
+    float* __restrict__ gb,           // [N] (atomicAdd +=)
+    int M, int N, int rows_per_block) {
+    const int col = blockIdx.x * 256 + threadIdx.x;
+    const int r0 = blockIdx.y * rows_per_block;
+    const int r1 = min(M, r0 + rows_per_block);
+    if (col >= N) return;
+    float s = 0.f;
+    for (int r = r0; r < r1; ++r) {
+        float v = bf2f(dY[(long)r * N + col]);
+        if constexpr (HAS_MASK) {
+            if (!(bf2f(mask[(long)r * N + col]) > 0.f)) v = 0.f;
+        }
+        s += v;
+    }
+    atomicAdd(&gb[col], s);
+}
+
+// ---------------------------------------------------------------- launchers
+
+extern "C" {
+
+void ss_gemm_nt(const void* A, const void* B, const void* bias,
+                const void* mask, void* C, int M, int N, int K,
+                bool relu, hipStream_t stream) {
+    dim3 block(256);
+    const bool has_bias = bias != nullptr;
+    const bool has_mask = mask != nullptr;
+
+    auto launch = [&](auto bm_tag, auto bias_tag, auto relu_tag, auto mask_tag) {
+        constexpr int BM = decltype(bm_tag)::value;
+        constexpr bool HB = decltype(bias_tag)::value;
+        constexpr bool RL = decltype(relu_tag)::value;
+        constexpr bool HM = decltype(mask_tag)::value;
+        constexpr int WAVES_M = BM == 64 ? 2 : 1;
+        constexpr int WAVES_N = BM == 64 ? 2 : 4;
+        dim3 grid(cdiv(M, BM), cdiv(N, 64));
+        hipLaunchKernelGGL(
+            (gemm_nt_kernel<BM, 64, WAVES_M, WAVES_N, HB, RL, HM>),
+            grid, block, 0, stream,
+            (const __bf16*)A, (const __bf16*)B, (const __bf16*)bias,
+            (const __bf16*)mask, (__bf16*)C, M, N, K);
+    };
+
+    using T = std::true_type;
+    using F = std::false_type;
+    using B64 = std::integral_constant<int, 64>;
+    using B32 = std::integral_constant<int, 32>;
+
+#define DISPATCH(BMT)                                                  \
+    if (has_bias && relu)        launch(BMT{}, T{}, T{}, F{});         \
+    else if (has_bias)           launch(BMT{}, T{}, F{}, F{});         \
+    else if (has_mask)           launch(BMT{}, F{}, F{}, T{});         \
+    else if (relu)               launch(BMT{}, F{}, T{}, F{});         \
+    else                         launch(BMT{}, F{}, F{}, F{});
+
+    if (M >= 48) { DISPATCH(B64) } else { DISPATCH(B32) }
+#undef DISPATCH
+}
+
+void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
+                 int Mo, int N, int Kb, int split_k, hipStream_t stream) {
+    // pick split so that grid covers the CUs: tiles * split >= ~512
+    if (split_k <= 0) {
+        const int tiles = cdiv(Mo, 64) * cdiv(N, 64);
+        split_k = 1;
+        while (tiles * split_k < 512 && split_k * 256 < Kb) split_k *= 2;
+    }
+    int k_per_split = cdiv(cdiv(Kb, split_k), BK) * BK;
+    split_k = cdiv(Kb, k_per_split);
+    dim3 grid(cdiv(Mo, 64), cdiv(N, 64), split_k);
+    dim3 block(256);
+    if (mask)
+        hipLaunchKernelGGL((wgrad_tn_kernel<true>), grid, block, 0, stream,
+                           (const __bf16*)dY, (const __bf16*)X,
+                           (const __bf16*)mask, (float*)gW, Mo, N, Kb,
+                           k_per_split);
+    else
+        hipLaunchKernelGGL((wgrad_tn_kernel<false>), grid, block, 0, stream,
+                           (const __bf16*)dY, (const __bf16*)X,
+                           (const __bf16*)mask, (float*)gW, Mo, N, Kb,
+                           k_per_split);
+}
+
+void ss_colsum(const void* dY, const void* mask, void* gb, int M, int N,
+               hipStream_t stream) {
+    int splits = M > 2048 ? 8 : 1;
+    int rows_per_block = cdiv(M, splits);
+    dim3 grid(cdiv(N, 256), splits);
+    dim3 block(256);
+    if (mask)
+        hipLaunchKernelGGL((colsum_kernel<true>), grid, block, 0, stream,
+                           (const __bf16*)dY, (const __bf16*)mask, (float*)gb,
+                           M, N, rows_per_block);
+    else
+        hipLaunchKernelGGL((colsum_kernel<false>), grid, block, 0, stream,
+                           (const __bf16*)dY, (const __bf16*)mask, (float*)gb,
+                           M, N, rows_per_block);
+}
+
+}  // extern "C"

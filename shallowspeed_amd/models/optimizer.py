@@ -18,22 +18,22 @@ class SGD:
         self.lr = float(lr)
         self._desc = None  # device-side descriptor table, built lazily
 
-    def _build_desc(self, ext):
+    def _build_desc(self):
         # Static descriptor: [master_ptr, grad_ptr, lp_ptr, lpt_ptr,
-        # numel, cols] per tensor, int64, lives on device.  Pointers are
-        # stable because parameter storage is persistent.
-        rows = []
+        # numel, cols, start] per tensor, int64, lives on device.
+        # Pointers are stable because parameter storage is persistent.
+        rows, start = [], 0
         for p in self.params:
             t = p.data
-            lp = p.lp
-            lpt = p.lp_t
-            cols = t.shape[1] if t.dim() == 2 else 0
+            cols = t.shape[1] if t.dim() == 2 else 1
             rows.append([
                 t.data_ptr(), p.grad.data_ptr(),
-                lp.data_ptr() if lp is not None else 0,
-                lpt.data_ptr() if lpt is not None else 0,
-                t.numel(), cols,
+                p.lp.data_ptr() if p.lp is not None else 0,
+                p.lp_t.data_ptr() if p.lp_t is not None else 0,
+                t.numel(), cols, start,
             ])
+            start += t.numel()
+        self._total = start
         cpu = torch.tensor(rows, dtype=torch.int64)
         self._desc = cpu.to(self.params[0].data.device)
         self._desc_cpu = cpu
@@ -44,8 +44,8 @@ class SGD:
         if self.params[0].data.is_cuda:
             ext = load_ext(required=True)
             if self._desc is None:
-                self._build_desc(ext)
-            ext.sgd_multi(self._desc, self.lr)
+                self._build_desc()
+            ext.sgd_multi(self._desc, self.lr, self._total)
         else:
             for p in self.params:
                 p.data -= self.lr * p.grad

@@ -1,0 +1,257 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (the GPU
+analog of the reference's finite-difference checks,
+tests/test_functional.py).  All tests need a real MI355X."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+BF = torch.bfloat16
+
+
+def ext():
+    from shallowspeed_amd.ops import load_ext
+
+    return load_ext(required=True)
+
+
+def rand_bf16(*shape, device, scale=1.0, seed=0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    t = torch.randn(*shape, generator=g) * scale
+    return t.to(device=device, dtype=BF)
+
+
+def tol(ref, atol=1e-2, rtol=2e-2):
+    return dict(atol=atol + 1e-3 * ref.abs().max().item(), rtol=rtol)
+
+
+# ------------------------------------------------------------- MFMA layout
+
+def test_mfma_layout(gpu_device):
+    """A=I with ASYMMETRIC B catches transposed C-writes
+    (cdna_hip_programming.md §3: always asymmetric-B check)."""
+    e = ext()
+    M = N = K = 64
+    a = torch.eye(M, K, device=gpu_device, dtype=BF)
+    b = torch.arange(N * K, device=gpu_device, dtype=torch.float32)
+    b = ((b % 37) / 37.0 + (b // K) * 0.01).reshape(N, K).to(BF)
+    empty = torch.Tensor()
+    c = e.gemm_nt(a, b, empty, empty, False)
+    torch.testing.assert_close(c.float(), b.t().float().contiguous(),
+                               atol=1e-2, rtol=1e-2)
+
+
+# ------------------------------------------------------------- gemm_nt
+
+@pytest.mark.parametrize("shape", [
+    (32, 10, 784), (64, 64, 64), (128, 128, 784), (100, 127, 123),
+    (8192, 128, 784), (33, 17, 9), (256, 10, 123),
+])
+def test_gemm_nt_vs_torch(gpu_device, shape):
+    M, N, K = shape
+    e = ext()
+    a = rand_bf16(M, K, device=gpu_device, seed=1)
+    b = rand_bf16(N, K, device=gpu_device, seed=2)
+    empty = torch.Tensor()
+    c = e.gemm_nt(a, b, empty, empty, False)
+    ref = a.float() @ b.float().t()
+    torch.testing.assert_close(c.float(), ref, **tol(ref))
+
+
+def test_gemm_nt_bias_relu(gpu_device):
+    e = ext()
+    M, N, K = 128, 96, 200
+    a = rand_bf16(M, K, device=gpu_device, seed=3)
+    b = rand_bf16(N, K, device=gpu_device, seed=4)
+    bias = rand_bf16(N, device=gpu_device, seed=5)
+    empty = torch.Tensor()
+    c = e.gemm_nt(a, b, bias, empty, True)
+    ref = torch.clamp(a.float() @ b.float().t() + bias.float(), min=0)
+    torch.testing.assert_close(c.float(), ref, **tol(ref))
+    assert (c.float() >= 0).all()
+
+
+def test_gemm_nt_mask(gpu_device):
+    """dgrad with fused ReLU mask: (a ⊙ 1[mask>0]) @ b^T."""
+    e = ext()
+    M, N, K = 96, 64, 128
+    a = rand_bf16(M, K, device=gpu_device, seed=6)
+    mask = rand_bf16(M, K, device=gpu_device, seed=7)
+    b = rand_bf16(N, K, device=gpu_device, seed=8)
+    empty = torch.Tensor()
+    c = e.gemm_nt(a, b, empty, mask, False)
+    am = a.float() * (mask.float() > 0)
+    ref = am @ b.float().t()
+    torch.testing.assert_close(c.float(), ref, **tol(ref))
+
+
+# ------------------------------------------------------------- wgrad
+
+@pytest.mark.parametrize("shape", [
+    (256, 64, 64), (8192, 128, 784), (100, 127, 123), (32, 10, 123),
+])
+def test_wgrad_accumulate_and_bias(gpu_device, shape):
+    Kb, Mo, N = shape
+    e = ext()
+    dy = rand_bf16(Kb, Mo, device=gpu_device, seed=9)
+    x = rand_bf16(Kb, N, device=gpu_device, seed=10)
+    gw = torch.ones(Mo, N, device=gpu_device, dtype=torch.float32)
+    gb = torch.ones(Mo, device=gpu_device, dtype=torch.float32)
+    empty = torch.Tensor()
+    e.wgrad_tn(dy, x, gw, gb, empty, 0)
+    ref_w = 1.0 + dy.float().t() @ x.float()
+    ref_b = 1.0 + dy.float().sum(0)
+    torch.testing.assert_close(gw, ref_w, **tol(ref_w))
+    torch.testing.assert_close(gb, ref_b, **tol(ref_b))
+
+
+def test_wgrad_mask_and_splitk_determinism_modes(gpu_device):
+    e = ext()
+    Kb, Mo, N = 4096, 64, 96
+    dy = rand_bf16(Kb, Mo, device=gpu_device, seed=11)
+    mask = rand_bf16(Kb, Mo, device=gpu_device, seed=12)
+    x = rand_bf16(Kb, N, device=gpu_device, seed=13)
+    dym = dy.float() * (mask.float() > 0)
+    ref_w = dym.t() @ x.float()
+    ref_b = dym.sum(0)
+    for split_k in (0, 1, 4):
+        gw = torch.zeros(Mo, N, device=gpu_device, dtype=torch.float32)
+        gb = torch.zeros(Mo, device=gpu_device, dtype=torch.float32)
+        e.wgrad_tn(dy, x, gw, gb, mask, split_k)
+        torch.testing.assert_close(gw, ref_w, **tol(ref_w))
+        torch.testing.assert_close(gb, ref_b, **tol(ref_b))
+
+
+# ------------------------------------------------------------- elementwise
+
+def test_relu_fwd_bwd(gpu_device):
+    e = ext()
+    x = rand_bf16(1000, 37, device=gpu_device, seed=14)
+    y = e.relu_fwd(x)
+    torch.testing.assert_close(y.float(), torch.clamp(x.float(), min=0))
+    dy = rand_bf16(1000, 37, device=gpu_device, seed=15)
+    dx = e.relu_bwd(dy, y)
+    torch.testing.assert_close(dx.float(), dy.float() * (y.float() > 0))
+
+
+@pytest.mark.parametrize("C", [10, 64, 100, 1000])
+def test_softmax_fwd_bwd(gpu_device, C):
+    e = ext()
+    x = rand_bf16(257, C, device=gpu_device, scale=3.0, seed=16)
+    s = e.softmax_fwd(x)
+    ref = torch.softmax(x.float(), dim=-1)
+    torch.testing.assert_close(s.float(), ref, atol=8e-3, rtol=2e-2)
+    dy = rand_bf16(257, C, device=gpu_device, seed=17)
+    dx = e.softmax_bwd(dy, s)
+    sf = s.float()
+    dot = (sf * dy.float()).sum(-1, keepdim=True)
+    torch.testing.assert_close(dx.float(), sf * (dy.float() - dot),
+                               atol=8e-3, rtol=2e-2)
+
+
+def test_head_bwd_kernels(gpu_device):
+    e = ext()
+    B, C, GB = 128, 10, 512
+    z = rand_bf16(B, C, device=gpu_device, scale=2.0, seed=18)
+    s = e.softmax_fwd(z)
+    t = torch.zeros(B, C)
+    t[torch.arange(B), torch.randint(0, C, (B,))] = 1
+    t = t.to(device=gpu_device, dtype=BF)
+    # mse head
+    sf = s.float()
+    g = -2.0 * (t.float() - sf) / GB
+    ref = sf * (g - (sf * g).sum(-1, keepdim=True))
+    dz = e.head_mse_bwd(s, t, GB)
+    torch.testing.assert_close(dz.float(), ref, atol=1e-4, rtol=2e-2)
+    # xent head
+    dz2 = e.head_xent_bwd(s, t, GB)
+    torch.testing.assert_close(dz2.float(), (sf - t.float()) / GB,
+                               atol=1e-4, rtol=2e-2)
+
+
+# ------------------------------------------------------------- SGD
+
+def test_sgd_multi(gpu_device):
+    from shallowspeed_amd.models import Linear, Sequential, SGD
+
+    model = Sequential([
+        Linear(123, 65, activation="relu"),
+        Linear(65, 10),
+    ]).materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.1)
+    for p in model.parameters():
+        p.grad.normal_(generator=None)
+    before = [p.data.clone() for p in model.parameters()]
+    grads = [p.grad.clone() for p in model.parameters()]
+    opt.step()
+    torch.cuda.synchronize()
+    for b, g, p in zip(before, grads, model.parameters()):
+        torch.testing.assert_close(p.data, b - 0.1 * g)
+        # bf16 copies refreshed in-kernel
+        torch.testing.assert_close(p.lp.float(), p.data.to(BF).float())
+        if p.lp_t is not None:
+            torch.testing.assert_close(p.lp_t.float(),
+                                       p.data.to(BF).t().float().contiguous())
+
+
+# ----------------------------------------------------- end-to-end on GPU
+
+def test_gpu_training_learns(gpu_device):
+    """Whole engine on HIP kernels: loss head + GEMMs + SGD learn the
+    synthetic teacher task (reference gate: accuracy climbs)."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import (
+        GPipeSchedule, NaiveParallelSchedule, PipeDreamFlushSchedule,
+        Topology, Worker)
+
+    model = MLP([784, 256, 128, 10], 0, 1, 512, loss="xent")
+    model.materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.05)
+    ds = Dataset(512, 128, n_samples=2048, in_dim=784, n_classes=10,
+                 device=gpu_device).load(0, 1)
+    w = Worker(Topology(device=gpu_device), model, ds, opt)
+
+    def acc():
+        model.eval()
+        p = model.forward(ds.x.to(BF), 0)
+        model.train()
+        return (p.float().argmax(-1).cpu() ==
+                ds.y.float().argmax(-1).cpu()).float().mean().item()
+
+    a0 = acc()
+    for sched in (NaiveParallelSchedule, GPipeSchedule,
+                  PipeDreamFlushSchedule):
+        for b in range(ds.num_batches()):
+            w.execute(sched(ds.num_mubatches(), 1, 0), b)
+    for _ in range(10):
+        for b in range(ds.num_batches()):
+            w.execute(GPipeSchedule(ds.num_mubatches(), 1, 0), b)
+    a1 = acc()
+    assert a1 > a0 + 0.15, (a0, a1)
+
+
+def test_gpu_matches_cpu_reference_one_step(gpu_device):
+    """One full training step GPU(bf16 HIP kernels) vs CPU(f32 torch):
+    weights agree to bf16 tolerance."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import NaiveParallelSchedule, Topology, Worker
+
+    def run(device):
+        model = MLP([48, 32, 16, 10], 0, 1, 64, loss="xent")
+        model.materialize_device(device)
+        opt = SGD(model.parameters(), lr=0.05)
+        ds = Dataset(64, 16, n_samples=64, in_dim=48, n_classes=10,
+                     device=device).load(0, 1)
+        w = Worker(Topology(device=torch.device(device)), model, ds, opt)
+        w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), 0)
+        if torch.device(device).type == "cuda":
+            torch.cuda.synchronize()
+        return [p.data.float().cpu() for p in model.parameters()]
+
+    got = run(gpu_device)
+    want = run("cpu")
+    for g, w_ in zip(got, want):
+        torch.testing.assert_close(g, w_, atol=3e-2, rtol=3e-2)

@@ -12,8 +12,8 @@
 extern "C" {
 void ss_gemm_nt(const void*, const void*, const void*, const void*, void*,
                 int, int, int, bool, hipStream_t);
-void ss_wgrad_tn(const void*, const void*, const void*, void*, int, int, int,
-                 int, hipStream_t);
+void ss_wgrad_tn(const void*, const void*, const void*, void*, void*, int,
+                 int, int, int, hipStream_t);
 void ss_colsum(const void*, const void*, void*, int, int, hipStream_t);
 void ss_relu_fwd(const void*, void*, long, hipStream_t);
 void ss_relu_bwd(const void*, const void*, void*, long, hipStream_t);
@@ -85,13 +85,31 @@ void wgrad_tn(torch::Tensor dy, torch::Tensor x, torch::Tensor gw,
         TORCH_CHECK(mask.sizes() == dy.sizes(), "mask must match dy");
         mask_p = mask.data_ptr();
     }
-    ss_wgrad_tn(dy.data_ptr(), x.data_ptr(), mask_p, gw.data_ptr(), Mo, N, Kb,
-                (int)split_k, cur_stream());
+    void* gb_p = nullptr;
     if (has(gb)) {
         check_f32(gb, "gb");
         TORCH_CHECK(gb.numel() == Mo, "gb size");
-        ss_colsum(dy.data_ptr(), mask_p, gb.data_ptr(), Kb, Mo, cur_stream());
+        gb_p = gb.data_ptr();
     }
+    // bias grad is FUSED into the wgrad kernel (n-tile-0 blocks sum
+    // the already-masked LDS dY tile) — no separate colsum launch or
+    // extra dY read on the hot path.
+    ss_wgrad_tn(dy.data_ptr(), x.data_ptr(), mask_p, gw.data_ptr(), gb_p, Mo,
+                N, Kb, (int)split_k, cur_stream());
+}
+
+torch::Tensor colsum(torch::Tensor dy, torch::Tensor mask) {
+    check_bf16(dy, "dy");
+    const void* mask_p = nullptr;
+    if (has(mask)) {
+        check_bf16(mask, "mask");
+        mask_p = mask.data_ptr();
+    }
+    auto gb = torch::zeros({dy.size(1)},
+                           dy.options().dtype(torch::kFloat));
+    ss_colsum(dy.data_ptr(), mask_p, gb.data_ptr(), dy.size(0), dy.size(1),
+              cur_stream());
+    return gb;
 }
 
 torch::Tensor relu_fwd(torch::Tensor x) {
@@ -166,7 +184,8 @@ void sgd_multi(torch::Tensor desc, double lr, int64_t total) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
-    m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum");
+    m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
+    m.def("colsum", &colsum, "standalone column sum (bias grad)");
     m.def("relu_fwd", &relu_fwd);
     m.def("relu_bwd", &relu_bwd);
     m.def("softmax_fwd", &softmax_fwd);

@@ -73,14 +73,27 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
             const int off = tid * EL;
             const int r = off / BK, c = off % BK;
             const int gm = m0 + r, gk = k0 + c;
-            if (gm < M && gk + EL <= K && !HAS_MASK) {
-                // fast path: contiguous vector load/store
+            if (gm < M && gk + EL <= K) {
+                // fast path: contiguous vector load (+ vector mask)
                 if constexpr (EL == 8) {
-                    *(ulonglong2*)&As[r][c] =
-                        *(const ulonglong2*)&A[(long)gm * K + gk];
+                    bf16x8 v = *(const bf16x8*)&A[(long)gm * K + gk];
+                    if constexpr (HAS_MASK) {
+                        bf16x8 mk = *(const bf16x8*)&mask[(long)gm * K + gk];
+#pragma unroll
+                        for (int i = 0; i < 8; ++i)
+                            if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+                    }
+                    *(bf16x8*)&As[r][c] = v;
                 } else {
-                    *(unsigned long long*)&As[r][c] =
-                        *(const unsigned long long*)&A[(long)gm * K + gk];
+                    using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
+                    bf16x4 v = *(const bf16x4*)&A[(long)gm * K + gk];
+                    if constexpr (HAS_MASK) {
+                        bf16x4 mk = *(const bf16x4*)&mask[(long)gm * K + gk];
+#pragma unroll
+                        for (int i = 0; i < 4; ++i)
+                            if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+                    }
+                    *(bf16x4*)&As[r][c] = v;
                 }
             } else {
 #pragma unroll
@@ -95,10 +108,6 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
                     }
                     As[r][c + i] = *(const ushort*)&v;
                 }
-            }
-            if constexpr (HAS_MASK) {
-                // mask path above is scalar; also apply on fast path
-                // (handled by taking the scalar branch when HAS_MASK)
             }
         }
         // ---- stage B tile [BN][BK] ----
@@ -173,6 +182,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const __bf16* __restrict__ X,     // [Kb][N]
     const __bf16* __restrict__ mask,  // [Kb][Mo]
     float* __restrict__ gW,           // [Mo][N] (atomicAdd +=)
+    float* __restrict__ gb,           // [Mo] or null: fused bias grad
     int Mo, int N, int Kb, int k_per_split) {
     constexpr int BM = 64, BN = 64;
     constexpr int LDA = BK + LDS_PAD;
@@ -198,6 +208,10 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 
     const int lrow = lane & 15;
     const int kch = lane >> 4;
+    // fused bias grad: only the n-tile-0 blocks compute it; thread
+    // (m = tid&63, kq = tid>>6) accumulates its quarter of each k-step
+    const bool do_db = (gb != nullptr) && (blockIdx.y == 0);
+    float db_part = 0.f;
 
     for (int k0 = kbeg; k0 < kend; k0 += BK) {
         // stage dY rows k0..k0+32 transposed into At[m][k]
@@ -206,18 +220,33 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             const int off = tid * EL;
             const int kr = off / BM, mc = off % BM;
             const int gk = k0 + kr;
+            if (gk < kend && m0 + BM <= Mo) {
+                bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + mc];
+                if constexpr (HAS_MASK) {
+                    bf16x8 mk = *(const bf16x8*)&mask[(long)gk * Mo + m0 + mc];
 #pragma unroll
-            for (int i = 0; i < EL; ++i) {
-                __bf16 v = (__bf16)0.f;
-                const int gm = m0 + mc + i;
-                if (gk < kend && gm < Mo) {
-                    v = dY[(long)gk * Mo + gm];
-                    if constexpr (HAS_MASK) {
-                        if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
-                            v = (__bf16)0.f;
-                    }
+                    for (int i = 0; i < 8; ++i)
+                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
                 }
-                At[mc + i][kr] = *(const ushort*)&v;
+#pragma unroll
+                for (int i = 0; i < EL; ++i) {
+                    const __bf16 e = v[i];
+                    At[mc + i][kr] = *(const ushort*)&e;
+                }
+            } else {
+#pragma unroll
+                for (int i = 0; i < EL; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    const int gm = m0 + mc + i;
+                    if (gk < kend && gm < Mo) {
+                        v = dY[(long)gk * Mo + gm];
+                        if constexpr (HAS_MASK) {
+                            if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
+                                v = (__bf16)0.f;
+                        }
+                    }
+                    At[mc + i][kr] = *(const ushort*)&v;
+                }
             }
         }
         // stage X rows transposed into Bt[n][k]
@@ -226,15 +255,31 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             const int off = tid * EL;
             const int kr = off / BN, nc = off % BN;
             const int gk = k0 + kr;
+            if (gk < kend && n0 + BN <= N) {
+                bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + nc];
 #pragma unroll
-            for (int i = 0; i < EL; ++i) {
-                __bf16 v = (__bf16)0.f;
-                const int gn = n0 + nc + i;
-                if (gk < kend && gn < N) v = X[(long)gk * N + gn];
-                Bt[nc + i][kr] = *(const ushort*)&v;
+                for (int i = 0; i < EL; ++i) {
+                    const __bf16 e = v[i];
+                    Bt[nc + i][kr] = *(const ushort*)&e;
+                }
+            } else {
+#pragma unroll
+                for (int i = 0; i < EL; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    const int gn = n0 + nc + i;
+                    if (gk < kend && gn < N) v = X[(long)gk * N + gn];
+                    Bt[nc + i][kr] = *(const ushort*)&v;
+                }
             }
         }
         __syncthreads();
+        if (do_db) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const ushort u = At[tid & 63][(tid >> 6) * 8 + j];
+                db_part += bf2f(*(const __bf16*)&u);
+            }
+        }
 
         bf16x8 a_frag[2], b_frag[2];
 #pragma unroll
@@ -264,6 +309,17 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                 if (grow >= Mo) continue;
                 atomicAdd(&gW[(long)grow * N + gcol], acc[i][j][r]);
             }
+        }
+    }
+
+    if (do_db) {
+        __shared__ float dbs[4][64];
+        dbs[tid >> 6][tid & 63] = db_part;
+        __syncthreads();
+        if (tid < 64 && m0 + tid < Mo) {
+            const float s4 = dbs[0][tid] + dbs[1][tid] + dbs[2][tid] +
+                             dbs[3][tid];
+            atomicAdd(&gb[m0 + tid], s4);
         }
     }
 }
@@ -334,7 +390,8 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
 }
 
 void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
-                 int Mo, int N, int Kb, int split_k, hipStream_t stream) {
+                 void* gb, int Mo, int N, int Kb, int split_k,
+                 hipStream_t stream) {
     // pick split so that grid covers the CUs: tiles * split >= ~512
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, 64) * cdiv(N, 64);
@@ -348,18 +405,22 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     if (mask)
         hipLaunchKernelGGL((wgrad_tn_kernel<true>), grid, block, 0, stream,
                            (const __bf16*)dY, (const __bf16*)X,
-                           (const __bf16*)mask, (float*)gW, Mo, N, Kb,
-                           k_per_split);
+                           (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,
+                           Kb, k_per_split);
     else
         hipLaunchKernelGGL((wgrad_tn_kernel<false>), grid, block, 0, stream,
                            (const __bf16*)dY, (const __bf16*)X,
-                           (const __bf16*)mask, (float*)gW, Mo, N, Kb,
-                           k_per_split);
+                           (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,
+                           Kb, k_per_split);
 }
 
 void ss_colsum(const void* dY, const void* mask, void* gb, int M, int N,
                hipStream_t stream) {
-    int splits = M > 2048 ? 8 : 1;
+    // enough blocks to cover the CUs (256 CUs; Guideline 11)
+    int col_tiles = cdiv(N, 256);
+    int splits = cdiv(512, col_tiles);
+    if (splits > cdiv(M, 64)) splits = cdiv(M, 64);
+    if (splits < 1) splits = 1;
     int rows_per_block = cdiv(M, splits);
     dim3 grid(cdiv(N, 256), splits);
     dim3 block(256);

@@ -163,10 +163,13 @@ class Linear(Module):
             self._stash("y", mubatch_id, y)
         return y
 
-    def backward(self, dout, mubatch_id: int = 0):
+    def backward(self, dout, mubatch_id: int = 0, need_dx: bool = True):
         x = self._unstash("x", mubatch_id)
         mask_src = self._unstash("y", mubatch_id) if self.activation == "relu" else None
-        dx = F.linear_dgrad(dout, self.weight.compute(), self.weight.compute_t(), mask_src)
+        dx = None
+        if need_dx:
+            dx = F.linear_dgrad(dout, self.weight.compute(),
+                                self.weight.compute_t(), mask_src)
         F.linear_wgrad_acc(dout, x, self.weight.grad, self.bias.grad, mask_src)
         return dx
 
@@ -261,6 +264,10 @@ class Sequential(Module):
         self.layers = list(layers)
         self._grad_hooks = []
         self._post_grad_hooks = []
+        # set True on pipeline stage 0: the first Linear's input grad is
+        # dL/d(data) — never consumed — so its dgrad GEMM is skipped
+        # (the most expensive dgrad: K = input width).
+        self._skip_input_grad = False
 
     def forward(self, inputs, mubatch_id: int = 0):
         x = inputs
@@ -270,8 +277,12 @@ class Sequential(Module):
 
     def backward(self, dout, mubatch_id: int = 0):
         d = dout
-        for layer in reversed(self.layers):
-            d = layer.backward(d, mubatch_id)
+        for i in range(len(self.layers) - 1, -1, -1):
+            layer = self.layers[i]
+            if i == 0 and self._skip_input_grad and isinstance(layer, Linear):
+                d = layer.backward(d, mubatch_id, need_dx=False)
+            else:
+                d = layer.backward(d, mubatch_id)
             for hook in self._grad_hooks:
                 for p in layer.parameters():
                     if p.requires_grad:
@@ -366,5 +377,6 @@ class MLP(Sequential):
             layers.append(head)
         super().__init__(layers)
         self.stage_idx, self.n_stages = stage_idx, n_stages
+        self._skip_input_grad = stage_idx == 0
         self.in_dim = bounds[0]
         self.out_dim = bounds[-1] if not is_last else sizes[-1]

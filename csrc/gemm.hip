@@ -28,6 +28,12 @@
 
 #define BK 32
 #define LDS_PAD 8
+// k-group XOR swizzle for transpose-staged tiles (wgrad): spreads the
+// per-instruction m-stride-8 scatter writes over banks (2-way instead
+// of 16-way) while fragment reads stay 16-B-aligned ds_read_b128.
+__device__ __forceinline__ int kswz(int m, int k) {
+    return ((((k) >> 3) ^ (((m) >> 3) & 3)) << 3) | ((k) & 7);
+}
 
 // ---------------------------------------------------------------- gemm_nt
 
@@ -231,7 +237,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #pragma unroll
                 for (int i = 0; i < EL; ++i) {
                     const __bf16 e = v[i];
-                    At[mc + i][kr] = *(const ushort*)&e;
+                    At[mc + i][kswz(mc + i, kr)] = *(const ushort*)&e;
                 }
             } else {
 #pragma unroll
@@ -245,7 +251,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                                 v = (__bf16)0.f;
                         }
                     }
-                    At[mc + i][kr] = *(const ushort*)&v;
+                    At[mc + i][kswz(mc + i, kr)] = *(const ushort*)&v;
                 }
             }
         }
@@ -260,7 +266,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #pragma unroll
                 for (int i = 0; i < EL; ++i) {
                     const __bf16 e = v[i];
-                    Bt[nc + i][kr] = *(const ushort*)&e;
+                    Bt[nc + i][kswz(nc + i, kr)] = *(const ushort*)&e;
                 }
             } else {
 #pragma unroll
@@ -268,26 +274,31 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                     __bf16 v = (__bf16)0.f;
                     const int gn = n0 + nc + i;
                     if (gk < kend && gn < N) v = X[(long)gk * N + gn];
-                    Bt[nc + i][kr] = *(const ushort*)&v;
+                    Bt[nc + i][kswz(nc + i, kr)] = *(const ushort*)&v;
                 }
             }
         }
         __syncthreads();
         if (do_db) {
+            const int m = tid & 63;
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                const ushort u = At[tid & 63][(tid >> 6) * 8 + j];
+                const ushort u = At[m][kswz(m, (tid >> 6) * 8 + j)];
                 db_part += bf2f(*(const __bf16*)&u);
             }
         }
 
         bf16x8 a_frag[2], b_frag[2];
 #pragma unroll
-        for (int i = 0; i < 2; ++i)
-            a_frag[i] = *(const bf16x8*)&At[wm * 32 + i * 16 + lrow][kch * 8];
+        for (int i = 0; i < 2; ++i) {
+            const int row = wm * 32 + i * 16 + lrow;
+            a_frag[i] = *(const bf16x8*)&At[row][kswz(row, kch * 8)];
+        }
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-            b_frag[j] = *(const bf16x8*)&Bt[wn * 32 + j * 16 + lrow][kch * 8];
+        for (int j = 0; j < 2; ++j) {
+            const int row = wn * 32 + j * 16 + lrow;
+            b_frag[j] = *(const bf16x8*)&Bt[row][kswz(row, kch * 8)];
+        }
 #pragma unroll
         for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -396,7 +407,7 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, 64) * cdiv(N, 64);
         split_k = 1;
-        while (tiles * split_k < 512 && split_k * 256 < Kb) split_k *= 2;
+        while (tiles * split_k < 1024 && split_k * 128 < Kb) split_k *= 2;
     }
     int k_per_split = cdiv(cdiv(Kb, split_k), BK) * BK;
     split_k = cdiv(Kb, k_per_split);

@@ -74,6 +74,17 @@ class Dataset:
         if self.device is not None:
             self.x = self.x.to(self.device)
             self.y = self.y.to(self.device)
+        # pre-cast compute copies on GPU: µbatch loads become straight
+        # D2D copies instead of cast+copy (saves an elementwise kernel
+        # + an allocation per µbatch on the hot path)
+        if self.device is not None and self.device.type == "cuda":
+            import torch as _t
+
+            self.x_compute = self.x.to(_t.bfloat16)
+            self.y_compute = self.y.to(_t.bfloat16)
+        else:
+            self.x_compute = self.x
+            self.y_compute = self.y
         return self
 
     def _load_full(self):
@@ -98,11 +109,11 @@ class Dataset:
 
     def micro_batch_input(self, batch_id: int, mubatch_id: int):
         lo, hi = self._mubatch_rows(batch_id, mubatch_id)
-        return self.x[lo:hi]
+        return self.x_compute[lo:hi]
 
     def micro_batch_target(self, batch_id: int, mubatch_id: int):
         lo, hi = self._mubatch_rows(batch_id, mubatch_id)
-        return self.y[lo:hi]
+        return self.y_compute[lo:hi]
 
     # counts — dataset.py:82-86
     def num_batches(self) -> int:

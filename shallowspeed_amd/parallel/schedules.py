@@ -67,6 +67,13 @@ class Schedule:
     def _buf(self, m):
         return m % (self.num_buffers // 2)
 
+    @property
+    def max_in_flight(self) -> int:
+        """Upper bound on µbatches with a live activation stash at any
+        point — 1 means the Worker may pass the shared input buffer to
+        forward without snapshotting it."""
+        return self.num_micro_batches
+
     # shared fragments ------------------------------------------------
     def _acquire_input(self, m):
         if self.is_first_stage:
@@ -107,6 +114,10 @@ class NaiveParallelSchedule(Schedule):
     @property
     def num_buffers(self):
         return 2
+
+    @property
+    def max_in_flight(self):
+        return 1  # fwd+bwd complete before the next µbatch starts
 
 
 class GPipeSchedule(Schedule):
@@ -201,6 +212,11 @@ class PipeDreamFlushSchedule(Schedule):
         # steady state with async sends
         return 4
 
+    @property
+    def max_in_flight(self):
+        return min(self.num_micro_batches,
+                   self.num_stages - self.stage_id) # warmup+1
+
 
 class InferenceSchedule(Schedule):
     """Forward-only pipeline for eval.  Reference: pipe.py:275-294
@@ -219,6 +235,10 @@ class InferenceSchedule(Schedule):
     @property
     def num_buffers(self):
         return 2
+
+    @property
+    def max_in_flight(self):
+        return 1  # eval stashes nothing
 
 
 SCHEDULES = {

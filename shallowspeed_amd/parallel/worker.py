@@ -103,6 +103,7 @@ class Worker:
     # --------------------------------------------------------- execution
     def execute(self, schedule, batch_id: int):
         self._batch_id = batch_id
+        self._snapshot_inputs = getattr(schedule, "max_in_flight", 2) > 1
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
@@ -182,12 +183,13 @@ class Worker:
 
     def _forward(self, cmd):
         x = self._in_bufs[cmd.in_buffer]
-        if self.model._training:
+        if self.model._training and self._snapshot_inputs:
             # Layers stash their input per µbatch for wgrad; the input
             # buffer is SHARED across µbatches (overwritten by the next
-            # Load/Recv), so snapshot it.  (The reference stashes the
-            # live buffer reference — layers.py:117 + pipe.py:447-454 —
-            # which is only safe for its naive schedule ordering.)
+            # Load/Recv), so snapshot it when more than one µbatch can
+            # be in flight.  (The reference stashes the live buffer
+            # reference — layers.py:117 + pipe.py:447-454 — which is
+            # only safe for its naive schedule ordering.)
             x = x.clone()
         y = self.model.forward(x, cmd.mubatch_id)
         self._wait_buffer("out", cmd.out_buffer)

@@ -32,6 +32,8 @@ def main():
     ap.add_argument("--layer-sizes", default="784,256,256,256,10")
     ap.add_argument("--loss", default="xent", choices=["xent", "mse"])
     ap.add_argument("--device", default=None)
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph step capture (single-GPU only)")
     args = ap.parse_args()
 
     from shallowspeed_amd.data import Dataset
@@ -69,8 +71,13 @@ def main():
     sched = sched_cls(ds.num_mubatches(), pp, topo.stage_id)
     nb = ds.num_batches()
 
+    use_graph = (not args.no_graph) and is_gpu and topo.world == 1
+
     def one_step(i):
-        worker.execute(sched, i % nb)
+        if use_graph:
+            worker.execute_graphed(sched, i % nb)
+        else:
+            worker.execute(sched, i % nb)
 
     import torch.distributed as dist
 
@@ -125,6 +132,7 @@ def main():
                 "schedule": args.schedule,
                 "loss": args.loss,
                 "mubatches": args.mubatches,
+                "hipgraph": use_graph,
             },
         }), flush=True)
 

@@ -124,6 +124,56 @@ class Worker:
     def enable_instruction_timing(self, on: bool = True):
         self._timing = on
 
+    # -------------------------------------------------- hipGraph replay
+    # The per-step kernel sequence is launch-bound at MLP scale
+    # (~20 small kernels/step); capturing it in a hipGraph collapses
+    # per-kernel host launch + boundary gaps into one replay
+    # (MI355X price list: eager host launch ≈3.3-3.8 µs/kernel vs
+    # graph replay ≈10-16 µs per whole step).
+    #
+    # Replay needs fixed addresses, so the batch's rows are first
+    # staged into persistent buffers OUTSIDE the graph, and the
+    # captured Load instructions read from the staging buffers.
+    def execute_graphed(self, schedule, batch_id: int):
+        if self.device.type != "cuda" or self.topo.world > 1:
+            return self.execute(schedule, batch_id)
+        ds = self.dataset
+        lb = ds.local_batch_size
+        if not hasattr(self, "_staged_x"):
+            self._staged_x = torch.empty_like(ds.x_compute[:lb])
+            self._staged_y = torch.empty_like(ds.y_compute[:lb])
+            self._graphs = {}
+        b0 = batch_id * lb
+        self._staged_x.copy_(ds.x_compute[b0:b0 + lb], non_blocking=True)
+        self._staged_y.copy_(ds.y_compute[b0:b0 + lb], non_blocking=True)
+        key = (type(schedule).__name__, schedule.num_micro_batches,
+               schedule.num_stages, schedule.stage_id, ds.mubatch_size)
+        g = self._graphs.get(key)
+        if g is None:
+            self._use_staged = True
+            try:
+                # eager warmup (allocates buffers, builds SGD desc) on a
+                # side stream, then capture
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    self.execute(schedule, batch_id)
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self.execute(schedule, batch_id)
+            except Exception:
+                self._use_staged = False
+                self._graphs[key] = False
+                return self.execute(schedule, batch_id)
+            finally:
+                self._use_staged = False
+            self._graphs[key] = g
+        elif g is False:  # capture failed before: stay eager
+            return self.execute(schedule, batch_id)
+        g.replay()
+
     # ----------------------------------------------------- async p2p
     # Sends are nonblocking isends (RCCL p2p over the direct xGMI link
     # between stage-adjacent GPUs); a buffer is only waited on when it
@@ -152,14 +202,22 @@ class Worker:
         self.optimizer.step()
 
     def _load_input(self, cmd):
-        x = self.dataset.micro_batch_input(self._batch_id, cmd.mubatch_id)
+        if getattr(self, "_use_staged", False):
+            mb = self.dataset.mubatch_size
+            x = self._staged_x[cmd.mubatch_id * mb:(cmd.mubatch_id + 1) * mb]
+        else:
+            x = self.dataset.micro_batch_input(self._batch_id, cmd.mubatch_id)
         self._wait_buffer("in", cmd.buffer_idx)
         buf = self._in_bufs[cmd.buffer_idx]
         assert x.shape == buf.shape, (x.shape, buf.shape)  # pipe.py:357-360
         buf.copy_(x.to(self.compute_dtype), non_blocking=True)
 
     def _load_target(self, cmd):
-        y = self.dataset.micro_batch_target(self._batch_id, cmd.mubatch_id)
+        if getattr(self, "_use_staged", False):
+            mb = self.dataset.mubatch_size
+            y = self._staged_y[cmd.mubatch_id * mb:(cmd.mubatch_id + 1) * mb]
+        else:
+            y = self.dataset.micro_batch_target(self._batch_id, cmd.mubatch_id)
         self._wait_buffer("gin", cmd.buffer_idx)
         buf = self._gin_bufs[cmd.buffer_idx]
         assert y.shape == buf.shape, (y.shape, buf.shape)  # pipe.py:362-365

@@ -139,18 +139,26 @@ class Worker:
             return self.execute(schedule, batch_id)
         ds = self.dataset
         lb = ds.local_batch_size
-        if not hasattr(self, "_staged_x"):
-            self._staged_x = torch.empty_like(ds.x_compute[:lb])
-            self._staged_y = torch.empty_like(ds.y_compute[:lb])
+        if not hasattr(self, "_graphs"):
             self._graphs = {}
-        b0 = batch_id * lb
-        self._staged_x.copy_(ds.x_compute[b0:b0 + lb], non_blocking=True)
-        self._staged_y.copy_(ds.y_compute[b0:b0 + lb], non_blocking=True)
+        # few batches: capture one graph PER batch_id (loads bake the
+        # batch's device addresses in — no staging copy per step).
+        # many batches: one graph reading persistent staging buffers,
+        # refilled with a single D2D copy per step.
+        per_batch = ds.num_batches() <= 8
         key = (type(schedule).__name__, schedule.num_micro_batches,
-               schedule.num_stages, schedule.stage_id, ds.mubatch_size)
+               schedule.num_stages, schedule.stage_id, ds.mubatch_size,
+               batch_id if per_batch else -1)
+        if not per_batch:
+            if not hasattr(self, "_staged_x"):
+                self._staged_x = torch.empty_like(ds.x_compute[:lb])
+                self._staged_y = torch.empty_like(ds.y_compute[:lb])
+            b0 = batch_id * lb
+            self._staged_x.copy_(ds.x_compute[b0:b0 + lb], non_blocking=True)
+            self._staged_y.copy_(ds.y_compute[b0:b0 + lb], non_blocking=True)
         g = self._graphs.get(key)
         if g is None:
-            self._use_staged = True
+            self._use_staged = not per_batch
             try:
                 # eager warmup (allocates buffers, builds SGD desc) on a
                 # side stream, then capture
@@ -164,7 +172,6 @@ class Worker:
                 with torch.cuda.graph(g):
                     self.execute(schedule, batch_id)
             except Exception:
-                self._use_staged = False
                 self._graphs[key] = False
                 return self.execute(schedule, batch_id)
             finally:

@@ -26,6 +26,8 @@
 
 #include "common.h"
 
+using bf16x4v = __attribute__((ext_vector_type(4))) __bf16;
+
 #define BK 32
 #define LDS_PAD 8
 // k-group XOR swizzle for transpose-staged tiles (wgrad): spreads the
@@ -190,11 +192,22 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     float* __restrict__ gW,           // [Mo][N] (atomicAdd +=)
     float* __restrict__ gb,           // [Mo] or null: fused bias grad
     int Mo, int N, int Kb, int k_per_split) {
+    // TN GEMM over the batch axis.  Tiles are staged in a BLOCKED
+    // [BK/4][BM/16][4][16] bf16 layout so that
+    //   * staging is plain 16-B vector writes (global rows are
+    //     m/n-contiguous — no software transpose), and
+    //   * MFMA A/B fragments come out of LDS through the gfx950
+    //     HARDWARE transpose read ds_read_b64_tr_b16
+    //     (__builtin_amdgcn_ds_read_tr16_b64_v4bf16): each 16-lane
+    //     group hands the crossbar one [4][16] subtile and receives
+    //     its column — semantics verified on hardware by
+    //     scripts/probe_tr.hip.
     constexpr int BM = 64, BN = 64;
-    constexpr int LDA = BK + LDS_PAD;
+    constexpr int NMB = BM / 16;  // m-blocks per tile
 
-    __shared__ ushort At[BM][LDA];  // dY^T slice: [m][k]
-    __shared__ ushort Bt[BN][LDA];  // X^T slice:  [n][k]
+    __shared__ ushort At[BK / 4 * NMB * 64];  // dY^T tile, blocked
+    __shared__ ushort Bt[BK / 4 * NMB * 64];  // X^T tile, blocked
+    __shared__ float dbs[4][64];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -212,70 +225,56 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const int lrow = lane & 15;
-    const int kch = lane >> 4;
-    // fused bias grad: only the n-tile-0 blocks compute it; thread
-    // (m = tid&63, kq = tid>>6) accumulates its quarter of each k-step
     const bool do_db = (gb != nullptr) && (blockIdx.y == 0);
     float db_part = 0.f;
 
+    // blocked-layout element address
+    auto baddr = [](int k, int m) {
+        return ((k >> 2) * NMB + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
+    };
+
+    // staging geometry: thread t covers (k = t/8, m = 8*(t%8) .. +8)
+    const int sk = tid >> 3;
+    const int sm = (tid & 7) * 8;
+
     for (int k0 = kbeg; k0 < kend; k0 += BK) {
-        // stage dY rows k0..k0+32 transposed into At[m][k]
-        {
-            constexpr int EL = 8;  // 32*64/256
-            const int off = tid * EL;
-            const int kr = off / BM, mc = off % BM;
-            const int gk = k0 + kr;
-            if (gk < kend && m0 + BM <= Mo) {
-                bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + mc];
-                if constexpr (HAS_MASK) {
-                    bf16x8 mk = *(const bf16x8*)&mask[(long)gk * Mo + m0 + mc];
+        const int gk = k0 + sk;
+        // ---- dY -> At ----
+        if (gk < kend && m0 + BM <= Mo) {
+            bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + sm];
+            if constexpr (HAS_MASK) {
+                bf16x8 mk = *(const bf16x8*)&mask[(long)gk * Mo + m0 + sm];
 #pragma unroll
-                    for (int i = 0; i < 8; ++i)
-                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                }
+                for (int i = 0; i < 8; ++i)
+                    if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+            }
+            *(bf16x8*)&At[baddr(sk, sm)] = v;
+        } else {
 #pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    const __bf16 e = v[i];
-                    At[mc + i][kswz(mc + i, kr)] = *(const ushort*)&e;
-                }
-            } else {
-#pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    __bf16 v = (__bf16)0.f;
-                    const int gm = m0 + mc + i;
-                    if (gk < kend && gm < Mo) {
-                        v = dY[(long)gk * Mo + gm];
-                        if constexpr (HAS_MASK) {
-                            if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
-                                v = (__bf16)0.f;
-                        }
+            for (int i = 0; i < 8; ++i) {
+                __bf16 v = (__bf16)0.f;
+                const int gm = m0 + sm + i;
+                if (gk < kend && gm < Mo) {
+                    v = dY[(long)gk * Mo + gm];
+                    if constexpr (HAS_MASK) {
+                        if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
+                            v = (__bf16)0.f;
                     }
-                    At[mc + i][kswz(mc + i, kr)] = *(const ushort*)&v;
                 }
+                At[baddr(sk, sm + i)] = *(const ushort*)&v;
             }
         }
-        // stage X rows transposed into Bt[n][k]
-        {
-            constexpr int EL = 8;
-            const int off = tid * EL;
-            const int kr = off / BN, nc = off % BN;
-            const int gk = k0 + kr;
-            if (gk < kend && n0 + BN <= N) {
-                bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + nc];
+        // ---- X -> Bt ----
+        if (gk < kend && n0 + BN <= N) {
+            bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + sm];
+            *(bf16x8*)&Bt[baddr(sk, sm)] = v;
+        } else {
 #pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    const __bf16 e = v[i];
-                    Bt[nc + i][kswz(nc + i, kr)] = *(const ushort*)&e;
-                }
-            } else {
-#pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    __bf16 v = (__bf16)0.f;
-                    const int gn = n0 + nc + i;
-                    if (gk < kend && gn < N) v = X[(long)gk * N + gn];
-                    Bt[nc + i][kswz(nc + i, kr)] = *(const ushort*)&v;
-                }
+            for (int i = 0; i < 8; ++i) {
+                __bf16 v = (__bf16)0.f;
+                const int gn = n0 + sm + i;
+                if (gk < kend && gn < N) v = X[(long)gk * N + gn];
+                Bt[baddr(sk, sm + i)] = *(const ushort*)&v;
             }
         }
         __syncthreads();
@@ -283,21 +282,41 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             const int m = tid & 63;
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                const ushort u = At[m][kswz(m, (tid >> 6) * 8 + j)];
+                const ushort u = At[baddr((tid >> 6) * 8 + j, m)];
                 db_part += bf2f(*(const __bf16*)&u);
             }
         }
 
+        // ---- fragments via hardware transpose read ----
+        typedef __attribute__((address_space(3))) bf16x4v* lds_v4p;
+        const int g2 = (lane >> 4) * 2;  // k-subtile pair for this group
+        const int lcol4 = (lane & 15) * 4;
         bf16x8 a_frag[2], b_frag[2];
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
-            const int row = wm * 32 + i * 16 + lrow;
-            a_frag[i] = *(const bf16x8*)&At[row][kswz(row, kch * 8)];
+            const int mblk = wm * 2 + i;
+            bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (lds_v4p)&At[(g2 * NMB + mblk) * 64 + lcol4]);
+            bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (lds_v4p)&At[((g2 + 1) * NMB + mblk) * 64 + lcol4]);
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                a_frag[i][e] = lo[e];
+                a_frag[i][e + 4] = hi[e];
+            }
         }
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
-            const int row = wn * 32 + j * 16 + lrow;
-            b_frag[j] = *(const bf16x8*)&Bt[row][kswz(row, kch * 8)];
+            const int nblk = wn * 2 + j;
+            bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
+            bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                b_frag[j][e] = lo[e];
+                b_frag[j][e + 4] = hi[e];
+            }
         }
 #pragma unroll
         for (int i = 0; i < 2; ++i)
@@ -308,6 +327,8 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
         __syncthreads();
     }
 
+    const int lrow = lane & 15;
+    const int kch = lane >> 4;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
 #pragma unroll
@@ -324,7 +345,6 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     }
 
     if (do_db) {
-        __shared__ float dbs[4][64];
         dbs[tid >> 6][tid & 63] = db_part;
         __syncthreads();
         if (tid < 64 && m0 + tid < Mo) {

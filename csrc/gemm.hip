@@ -39,6 +39,59 @@ __device__ __forceinline__ int kswz(int m, int k) {
 
 // ---------------------------------------------------------------- gemm_nt
 
+// stage a [TILE_ROWS][BK] row-major slice (rows base row0, k base k0)
+// into a padded LDS tile; zero-fills out-of-range; optional >0 mask
+// applied elementwise (fused ReLU backward).
+template <int TILE_ROWS, bool HAS_MASK>
+__device__ __forceinline__ void stage_nt_tile(
+    ushort (*__restrict__ dst)[BK + LDS_PAD],
+    const __bf16* __restrict__ src, const __bf16* __restrict__ msk,
+    int nrows, int K, int row0, int k0, int tid) {
+    constexpr int EL = TILE_ROWS * BK / 256;  // 4, 8 or 16 per thread
+    constexpr int V = EL < 8 ? 4 : 8;
+    const int off = tid * EL;
+    const int r = off / BK, c = off % BK;
+    const int g = row0 + r, gk = k0 + c;
+    if (g < nrows && gk + EL <= K) {
+#pragma unroll
+        for (int ch = 0; ch < EL; ch += V) {
+            if constexpr (V == 8) {
+                bf16x8 v = *(const bf16x8*)&src[(long)g * K + gk + ch];
+                if constexpr (HAS_MASK) {
+                    bf16x8 mk = *(const bf16x8*)&msk[(long)g * K + gk + ch];
+#pragma unroll
+                    for (int i = 0; i < 8; ++i)
+                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+                }
+                *(bf16x8*)&dst[r][c + ch] = v;
+            } else {
+                bf16x4v v = *(const bf16x4v*)&src[(long)g * K + gk + ch];
+                if constexpr (HAS_MASK) {
+                    bf16x4v mk = *(const bf16x4v*)&msk[(long)g * K + gk + ch];
+#pragma unroll
+                    for (int i = 0; i < 4; ++i)
+                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+                }
+                *(bf16x4v*)&dst[r][c + ch] = v;
+            }
+        }
+    } else {
+#pragma unroll
+        for (int i = 0; i < EL; ++i) {
+            __bf16 v = (__bf16)0.f;
+            if (g < nrows && gk + i < K) {
+                v = src[(long)g * K + gk + i];
+                if constexpr (HAS_MASK) {
+                    if (!(bf2f(msk[(long)g * K + gk + i]) > 0.f))
+                        v = (__bf16)0.f;
+                }
+            }
+            dst[r][c + i] = *(const ushort*)&v;
+        }
+    }
+}
+
+
 template <int BM, int BN, int WAVES_M, int WAVES_N,
           bool HAS_BIAS, bool RELU, bool HAS_MASK>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
@@ -75,72 +128,9 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const int kch = lane >> 4;  // 0..3 -> k offset kch*8
 
     for (int k0 = 0; k0 < K; k0 += BK) {
-        // ---- stage A tile [BM][BK] (row-major slice of A) ----
-        {
-            constexpr int EL = BM * BK / 256;  // elems per thread (4 or 8)
-            const int off = tid * EL;
-            const int r = off / BK, c = off % BK;
-            const int gm = m0 + r, gk = k0 + c;
-            if (gm < M && gk + EL <= K) {
-                // fast path: contiguous vector load (+ vector mask)
-                if constexpr (EL == 8) {
-                    bf16x8 v = *(const bf16x8*)&A[(long)gm * K + gk];
-                    if constexpr (HAS_MASK) {
-                        bf16x8 mk = *(const bf16x8*)&mask[(long)gm * K + gk];
-#pragma unroll
-                        for (int i = 0; i < 8; ++i)
-                            if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                    }
-                    *(bf16x8*)&As[r][c] = v;
-                } else {
-                    using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
-                    bf16x4 v = *(const bf16x4*)&A[(long)gm * K + gk];
-                    if constexpr (HAS_MASK) {
-                        bf16x4 mk = *(const bf16x4*)&mask[(long)gm * K + gk];
-#pragma unroll
-                        for (int i = 0; i < 4; ++i)
-                            if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                    }
-                    *(bf16x4*)&As[r][c] = v;
-                }
-            } else {
-#pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    __bf16 v = (__bf16)0.f;
-                    if (gm < M && gk + i < K) {
-                        v = A[(long)gm * K + gk + i];
-                        if constexpr (HAS_MASK) {
-                            if (!(bf2f(mask[(long)gm * K + gk + i]) > 0.f))
-                                v = (__bf16)0.f;
-                        }
-                    }
-                    As[r][c + i] = *(const ushort*)&v;
-                }
-            }
-        }
-        // ---- stage B tile [BN][BK] ----
-        {
-            constexpr int EL = BN * BK / 256;
-            const int off = tid * EL;
-            const int r = off / BK, c = off % BK;
-            const int gn = n0 + r, gk = k0 + c;
-            if (gn < N && gk + EL <= K) {
-                if constexpr (EL == 8) {
-                    *(ulonglong2*)&Bs[r][c] =
-                        *(const ulonglong2*)&B[(long)gn * K + gk];
-                } else {
-                    *(unsigned long long*)&Bs[r][c] =
-                        *(const unsigned long long*)&B[(long)gn * K + gk];
-                }
-            } else {
-#pragma unroll
-                for (int i = 0; i < EL; ++i) {
-                    __bf16 v = (__bf16)0.f;
-                    if (gn < N && gk + i < K) v = B[(long)gn * K + gk + i];
-                    Bs[r][c + i] = *(const ushort*)&v;
-                }
-            }
-        }
+        // ---- stage A tile [BM][BK] and B tile [BN][BK] ----
+        stage_nt_tile<BM, HAS_MASK>(As, A, mask, M, K, m0, k0, tid);
+        stage_nt_tile<BN, false>(Bs, B, nullptr, N, K, n0, k0, tid);
         __syncthreads();
 
         // ---- MFMA ----
@@ -203,10 +193,11 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     //     its column — semantics verified on hardware by
     //     scripts/probe_tr.hip.
     constexpr int BM = 64, BN = 64;
+    constexpr int BKW = 64;       // K-step (batch rows per stage)
     constexpr int NMB = BM / 16;  // m-blocks per tile
 
-    __shared__ ushort At[BK / 4 * NMB * 64];  // dY^T tile, blocked
-    __shared__ ushort Bt[BK / 4 * NMB * 64];  // X^T tile, blocked
+    __shared__ ushort At[BKW / 4 * NMB * 64];  // dY^T tile, blocked
+    __shared__ ushort Bt[BKW / 4 * NMB * 64];  // X^T tile, blocked
     __shared__ float dbs[4][64];
 
     const int tid = threadIdx.x;
@@ -233,25 +224,29 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
         return ((k >> 2) * NMB + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
     };
 
-    // staging geometry: thread t covers (k = t/8, m = 8*(t%8) .. +8)
-    const int sk = tid >> 3;
-    const int sm = (tid & 7) * 8;
+    // staging geometry: thread t covers (k = t/4, m = 16*(t%4) .. +16)
+    const int sk = tid >> 2;
+    const int sm = (tid & 3) * 16;
 
-    for (int k0 = kbeg; k0 < kend; k0 += BK) {
+    for (int k0 = kbeg; k0 < kend; k0 += BKW) {
         const int gk = k0 + sk;
         // ---- dY -> At ----
         if (gk < kend && m0 + BM <= Mo) {
-            bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + sm];
-            if constexpr (HAS_MASK) {
-                bf16x8 mk = *(const bf16x8*)&mask[(long)gk * Mo + m0 + sm];
 #pragma unroll
-                for (int i = 0; i < 8; ++i)
-                    if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+            for (int ch = 0; ch < 16; ch += 8) {
+                bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + sm + ch];
+                if constexpr (HAS_MASK) {
+                    bf16x8 mk =
+                        *(const bf16x8*)&mask[(long)gk * Mo + m0 + sm + ch];
+#pragma unroll
+                    for (int i = 0; i < 8; ++i)
+                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
+                }
+                *(bf16x8*)&At[baddr(sk, sm + ch)] = v;
             }
-            *(bf16x8*)&At[baddr(sk, sm)] = v;
         } else {
 #pragma unroll
-            for (int i = 0; i < 8; ++i) {
+            for (int i = 0; i < 16; ++i) {
                 __bf16 v = (__bf16)0.f;
                 const int gm = m0 + sm + i;
                 if (gk < kend && gm < Mo) {
@@ -266,11 +261,14 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
         }
         // ---- X -> Bt ----
         if (gk < kend && n0 + BN <= N) {
-            bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + sm];
-            *(bf16x8*)&Bt[baddr(sk, sm)] = v;
+#pragma unroll
+            for (int ch = 0; ch < 16; ch += 8) {
+                bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + sm + ch];
+                *(bf16x8*)&Bt[baddr(sk, sm + ch)] = v;
+            }
         } else {
 #pragma unroll
-            for (int i = 0; i < 8; ++i) {
+            for (int i = 0; i < 16; ++i) {
                 __bf16 v = (__bf16)0.f;
                 const int gn = n0 + sm + i;
                 if (gk < kend && gn < N) v = X[(long)gk * N + gn];
@@ -281,49 +279,52 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
         if (do_db) {
             const int m = tid & 63;
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const ushort u = At[baddr((tid >> 6) * 8 + j, m)];
+            for (int j = 0; j < 16; ++j) {
+                const ushort u = At[baddr((tid >> 6) * 16 + j, m)];
                 db_part += bf2f(*(const __bf16*)&u);
             }
         }
 
         // ---- fragments via hardware transpose read ----
         typedef __attribute__((address_space(3))) bf16x4v* lds_v4p;
-        const int g2 = (lane >> 4) * 2;  // k-subtile pair for this group
         const int lcol4 = (lane & 15) * 4;
-        bf16x8 a_frag[2], b_frag[2];
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
-            const int mblk = wm * 2 + i;
-            bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (lds_v4p)&At[(g2 * NMB + mblk) * 64 + lcol4]);
-            bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (lds_v4p)&At[((g2 + 1) * NMB + mblk) * 64 + lcol4]);
+        for (int kk = 0; kk < BKW / 32; ++kk) {
+            const int g2 = kk * 8 + (lane >> 4) * 2;  // k-subtile pair
+            bf16x8 a_frag[2], b_frag[2];
 #pragma unroll
-            for (int e = 0; e < 4; ++e) {
-                a_frag[i][e] = lo[e];
-                a_frag[i][e + 4] = hi[e];
+            for (int i = 0; i < 2; ++i) {
+                const int mblk = wm * 2 + i;
+                bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (lds_v4p)&At[(g2 * NMB + mblk) * 64 + lcol4]);
+                bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (lds_v4p)&At[((g2 + 1) * NMB + mblk) * 64 + lcol4]);
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    a_frag[i][e] = lo[e];
+                    a_frag[i][e + 4] = hi[e];
+                }
             }
-        }
 #pragma unroll
-        for (int j = 0; j < 2; ++j) {
-            const int nblk = wn * 2 + j;
-            bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
-            bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
+            for (int j = 0; j < 2; ++j) {
+                const int nblk = wn * 2 + j;
+                bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
+                bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
 #pragma unroll
-            for (int e = 0; e < 4; ++e) {
-                b_frag[j][e] = lo[e];
-                b_frag[j][e + 4] = hi[e];
+                for (int e = 0; e < 4; ++e) {
+                    b_frag[j][e] = lo[e];
+                    b_frag[j][e + 4] = hi[e];
+                }
             }
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
         }
-#pragma unroll
-        for (int i = 0; i < 2; ++i)
-#pragma unroll
-            for (int j = 0; j < 2; ++j)
-                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
         __syncthreads();
     }
 
@@ -394,11 +395,12 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
         constexpr bool HB = decltype(bias_tag)::value;
         constexpr bool RL = decltype(relu_tag)::value;
         constexpr bool HM = decltype(mask_tag)::value;
-        constexpr int WAVES_M = BM == 64 ? 2 : 1;
-        constexpr int WAVES_N = BM == 64 ? 2 : 4;
-        dim3 grid(cdiv(M, BM), cdiv(N, 64));
+        constexpr int BN = BM == 128 ? 128 : 64;
+        constexpr int WAVES_M = BM == 32 ? 1 : 2;
+        constexpr int WAVES_N = BM == 32 ? 4 : 2;
+        dim3 grid(cdiv(M, BM), cdiv(N, BN));
         hipLaunchKernelGGL(
-            (gemm_nt_kernel<BM, 64, WAVES_M, WAVES_N, HB, RL, HM>),
+            (gemm_nt_kernel<BM, BN, WAVES_M, WAVES_N, HB, RL, HM>),
             grid, block, 0, stream,
             (const __bf16*)A, (const __bf16*)B, (const __bf16*)bias,
             (const __bf16*)mask, (__bf16*)C, M, N, K);
@@ -406,6 +408,7 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
 
     using T = std::true_type;
     using F = std::false_type;
+    using B128 = std::integral_constant<int, 128>;
     using B64 = std::integral_constant<int, 64>;
     using B32 = std::integral_constant<int, 32>;
 
@@ -416,7 +419,9 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
     else if (relu)               launch(BMT{}, F{}, T{}, F{});         \
     else                         launch(BMT{}, F{}, F{}, F{});
 
-    if (M >= 48) { DISPATCH(B64) } else { DISPATCH(B32) }
+    if (M >= 4096 && N >= 128) { DISPATCH(B128) }
+    else if (M >= 48) { DISPATCH(B64) }
+    else { DISPATCH(B32) }
 #undef DISPATCH
 }
 
@@ -429,7 +434,7 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
         split_k = 1;
         while (tiles * split_k < 1024 && split_k * 128 < Kb) split_k *= 2;
     }
-    int k_per_split = cdiv(cdiv(Kb, split_k), BK) * BK;
+    int k_per_split = cdiv(cdiv(Kb, split_k), 64) * 64;  // wgrad BKW=64
     split_k = cdiv(Kb, k_per_split);
     dim3 grid(cdiv(Mo, 64), cdiv(N, 64), split_k);
     dim3 block(256);

@@ -419,9 +419,15 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
     else if (relu)               launch(BMT{}, F{}, T{}, F{});         \
     else                         launch(BMT{}, F{}, F{}, F{});
 
-    if (M >= 4096 && N >= 128) { DISPATCH(B128) }
-    else if (M >= 48) { DISPATCH(B64) }
-    else { DISPATCH(B32) }
+    // 128x128 only when the grid still covers the CUs with >=2
+    // blocks each (1 block/CU = 4 waves starves latency hiding)
+    if (N >= 128 && (long)cdiv(M, 128) * cdiv(N, 128) >= 512) {
+        DISPATCH(B128)
+    } else if (M >= 48) {
+        DISPATCH(B64)
+    } else {
+        DISPATCH(B32)
+    }
 #undef DISPATCH
 }
 

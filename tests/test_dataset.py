@@ -36,3 +36,19 @@ def test_mubatch_slicing_arithmetic():
 def test_targets_one_hot():
     ds = Dataset(16, 4, n_samples=32).load(0, 1)
     torch.testing.assert_close(ds.y.sum(-1), torch.ones(32))
+
+
+def test_file_backed_load(tmp_path):
+    """Reference on-disk format: x_train.parquet + y_train.npy
+    (download_dataset.py:10-23 analog written by prepare_data.py)."""
+    import subprocess
+    import sys
+
+    subprocess.run([sys.executable, "prepare_data.py", "--out",
+                    str(tmp_path), "--samples", "1000"], check=True)
+    ds = Dataset(32, 8, save_dir=str(tmp_path), n_samples=5).load(0, 1)
+    assert ds.x.shape[0] >= 800 and ds.x.shape[1] == 784  # 85% of 1000
+    assert ds.y.shape[1] == 10
+    val = Dataset(32, 8, save_dir=str(tmp_path), validation=True,
+                  n_samples=5).load(0, 1)
+    assert val.x.shape[0] <= 160

@@ -56,7 +56,8 @@ def _run_dist(fn, world, tmp_path, **kwargs):
 
 # ------------------------------------------------------------------ DP
 
-def _dp_train(rank, world, out_dir, loss="mse", schedule="naive"):
+def _dp_train(rank, world, out_dir, loss="mse", schedule="naive",
+              bucket_bytes=25 << 20):
     from shallowspeed_amd.parallel import init_topology
 
     topo = init_topology(dp=world, pp=1, backend="gloo",
@@ -67,7 +68,9 @@ def _dp_train(rank, world, out_dir, loss="mse", schedule="naive"):
     ds = Dataset(GBS, (GBS // world) // MUB,
                  n_samples=N, in_dim=SIZES[0], n_classes=SIZES[-1])
     ds.load(topo.dp_rank, world)
-    w = Worker(topo, model, ds, opt)
+    w = Worker(topo, model, ds, opt, bucket_bytes=bucket_bytes)
+    if bucket_bytes < (1 << 20):
+        assert len(w.reducer.buckets) > 1, "test wants multiple buckets"
     cls = SCHEDULES[schedule]
     for b in range(ds.num_batches()):
         w.execute(cls(ds.num_mubatches(), 1, 0), b)
@@ -85,6 +88,16 @@ def test_dp2_matches_serial(tmp_path, schedule):
     got = torch.load(tmp_path / "dp_params.pt", weights_only=False)
     want = _serial_params()
     assert len(got) == len(want)
+    for g, w in zip(got, want):
+        torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)
+
+
+def test_dp2_multibucket_matches_serial(tmp_path):
+    """Small bucket size => several overlapped all-reduces per backward
+    (the bucketed path with >1 bucket)."""
+    _run_dist(_dp_train, 2, tmp_path, schedule="gpipe", bucket_bytes=2048)
+    got = torch.load(tmp_path / "dp_params.pt", weights_only=False)
+    want = _serial_params()
     for g, w in zip(got, want):
         torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)
 

@@ -166,29 +166,6 @@ class Linear(Module):
     def backward(self, dout, mubatch_id: int = 0, need_dx: bool = True):
         x = self._unstash("x", mubatch_id)
         mask_src = self._unstash("y", mubatch_id) if self.activation == "relu" else None
-        if dout.is_cuda:
-            # wgrad is off the critical path: run it on the side stream,
-            # overlapping the dgrad chain.  Mark the inputs as used by
-            # the side stream so the caching allocator keeps them alive.
-            s = F.wgrad_stream()
-            ev = torch.cuda.Event()
-            ev.record()
-            s.wait_event(ev)
-            with torch.cuda.stream(s):
-                F.linear_wgrad_acc(dout, x, self.weight.grad,
-                                   self.bias.grad, mask_src)
-                for t in (dout, x, mask_src):
-                    if t is not None:
-                        t.record_stream(s)
-            done = torch.cuda.Event()
-            done.record(s)
-            self.weight._grad_event = done
-            self.bias._grad_event = done
-            dx = None
-            if need_dx:
-                dx = F.linear_dgrad(dout, self.weight.compute(),
-                                    self.weight.compute_t(), mask_src)
-            return dx
         dx = None
         if need_dx:
             dx = F.linear_dgrad(dout, self.weight.compute(),

@@ -45,11 +45,6 @@ class SGD:
             ext = load_ext(required=True)
             if self._desc is None:
                 self._build_desc()
-            # grads are produced on the side wgrad stream — join it
-            from ..ops.functional import _WGRAD_STREAM
-
-            if _WGRAD_STREAM is not None:
-                torch.cuda.current_stream().wait_stream(_WGRAD_STREAM)
             ext.sgd_multi(self._desc, self.lr, self._total)
         else:
             for p in self.params:

@@ -31,23 +31,6 @@ def _ext_for(t: torch.Tensor):
     return load_ext(required=True) if _is_gpu(t) else None
 
 
-_WGRAD_STREAM = None
-
-
-def wgrad_stream():
-    """Side HIP stream for weight-gradient kernels.
-
-    wgrad/bias-grad are OFF the backward critical path (only the
-    optimizer consumes them), so they run on this stream and overlap
-    the dgrad chain on the main stream.  Consumers (fused SGD, DP
-    bucket all-reduce) synchronize via the per-parameter events the
-    layer records (Parameter._grad_event)."""
-    global _WGRAD_STREAM
-    if _WGRAD_STREAM is None:
-        _WGRAD_STREAM = torch.cuda.Stream()
-    return _WGRAD_STREAM
-
-
 # ---------------------------------------------------------------- linear
 
 def linear_fwd(x, w, b=None, relu=False):

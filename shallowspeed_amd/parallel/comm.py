@@ -191,8 +191,7 @@ class GradReducer:
 
     def _push_bucket(self, lo, hi, params):
         bid = len(self.buckets)
-        self.buckets.append({"lo": lo, "hi": hi, "nparams": len(params),
-                             "params": list(params)})
+        self.buckets.append({"lo": lo, "hi": hi, "nparams": len(params)})
         for p in params:
             self.param_bucket[id(p)] = bid
 
@@ -209,15 +208,6 @@ class GradReducer:
         self._pending[bid] -= 1
         if self._pending[bid] == 0:
             b = self.buckets[bid]
-            if self.flat.is_cuda:
-                # grads land on the side wgrad stream: the collective
-                # (enqueued w.r.t. the current stream) must wait for
-                # this bucket's wgrad kernels
-                cur = torch.cuda.current_stream()
-                for p in b["params"]:
-                    ev = getattr(p, "_grad_event", None)
-                    if ev is not None:
-                        cur.wait_event(ev)
             h = dist.all_reduce(
                 self.flat[b["lo"]:b["hi"]], op=dist.ReduceOp.SUM,
                 group=self.group, async_op=True,

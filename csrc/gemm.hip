@@ -39,58 +39,86 @@ __device__ __forceinline__ int kswz(int m, int k) {
 
 // ---------------------------------------------------------------- gemm_nt
 
-// stage a [TILE_ROWS][BK] row-major slice (rows base row0, k base k0)
-// into a padded LDS tile; zero-fills out-of-range; optional >0 mask
-// applied elementwise (fused ReLU backward).
+// Staging is split into an ISSUE half (global -> registers, with
+// bounds zero-fill and the optional fused >0 mask) and a WRITE half
+// (registers -> padded LDS tile) so the HBM latency of tile t+1 hides
+// under tile t's MFMA work (async-STAGE split, one barrier per K-step;
+// cdna_hip_programming.md Guideline 15 / T14).
 template <int TILE_ROWS, bool HAS_MASK>
-__device__ __forceinline__ void stage_nt_tile(
-    ushort (*__restrict__ dst)[BK + LDS_PAD],
-    const __bf16* __restrict__ src, const __bf16* __restrict__ msk,
-    int nrows, int K, int row0, int k0, int tid) {
-    constexpr int EL = TILE_ROWS * BK / 256;  // 4, 8 or 16 per thread
-    constexpr int V = EL < 8 ? 4 : 8;
-    const int off = tid * EL;
-    const int r = off / BK, c = off % BK;
-    const int g = row0 + r, gk = k0 + c;
-    if (g < nrows && gk + EL <= K) {
+struct StageReg {
+    static constexpr int EL = TILE_ROWS * BK / 256;  // 4, 8 or 16
+    __bf16 v[EL];
+
+    __device__ __forceinline__ void load(const __bf16* __restrict__ src,
+                                         const __bf16* __restrict__ msk,
+                                         int nrows, int K, int row0, int k0,
+                                         int tid) {
+        const int off = tid * EL;
+        const int r = off / BK, c = off % BK;
+        const int g = row0 + r, gk = k0 + c;
+        if (g < nrows && gk + EL <= K) {
 #pragma unroll
-        for (int ch = 0; ch < EL; ch += V) {
-            if constexpr (V == 8) {
-                bf16x8 v = *(const bf16x8*)&src[(long)g * K + gk + ch];
-                if constexpr (HAS_MASK) {
-                    bf16x8 mk = *(const bf16x8*)&msk[(long)g * K + gk + ch];
+            for (int ch = 0; ch < EL; ch += (EL < 8 ? 4 : 8)) {
+                if constexpr (EL >= 8) {
+                    bf16x8 t = *(const bf16x8*)&src[(long)g * K + gk + ch];
+                    if constexpr (HAS_MASK) {
+                        bf16x8 mk =
+                            *(const bf16x8*)&msk[(long)g * K + gk + ch];
 #pragma unroll
-                    for (int i = 0; i < 8; ++i)
-                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                }
-                *(bf16x8*)&dst[r][c + ch] = v;
-            } else {
-                bf16x4v v = *(const bf16x4v*)&src[(long)g * K + gk + ch];
-                if constexpr (HAS_MASK) {
-                    bf16x4v mk = *(const bf16x4v*)&msk[(long)g * K + gk + ch];
+                        for (int i = 0; i < 8; ++i)
+                            if (!(bf2f(mk[i]) > 0.f)) t[i] = (__bf16)0.f;
+                    }
 #pragma unroll
-                    for (int i = 0; i < 4; ++i)
-                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                }
-                *(bf16x4v*)&dst[r][c + ch] = v;
-            }
-        }
-    } else {
+                    for (int i = 0; i < 8; ++i) v[ch + i] = t[i];
+                } else {
+                    bf16x4v t = *(const bf16x4v*)&src[(long)g * K + gk + ch];
+                    if constexpr (HAS_MASK) {
+                        bf16x4v mk =
+                            *(const bf16x4v*)&msk[(long)g * K + gk + ch];
 #pragma unroll
-        for (int i = 0; i < EL; ++i) {
-            __bf16 v = (__bf16)0.f;
-            if (g < nrows && gk + i < K) {
-                v = src[(long)g * K + gk + i];
-                if constexpr (HAS_MASK) {
-                    if (!(bf2f(msk[(long)g * K + gk + i]) > 0.f))
-                        v = (__bf16)0.f;
+                        for (int i = 0; i < 4; ++i)
+                            if (!(bf2f(mk[i]) > 0.f)) t[i] = (__bf16)0.f;
+                    }
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) v[ch + i] = t[i];
                 }
             }
-            dst[r][c + i] = *(const ushort*)&v;
+        } else {
+#pragma unroll
+            for (int i = 0; i < EL; ++i) {
+                __bf16 t = (__bf16)0.f;
+                if (g < nrows && gk + i < K) {
+                    t = src[(long)g * K + gk + i];
+                    if constexpr (HAS_MASK) {
+                        if (!(bf2f(msk[(long)g * K + gk + i]) > 0.f))
+                            t = (__bf16)0.f;
+                    }
+                }
+                v[i] = t;
+            }
         }
     }
-}
 
+    __device__ __forceinline__ void write(
+        ushort (*__restrict__ dst)[BK + LDS_PAD], int tid) const {
+        const int off = tid * EL;
+        const int r = off / BK, c = off % BK;
+#pragma unroll
+        for (int ch = 0; ch < EL; ch += (EL < 8 ? 4 : 8)) {
+            if constexpr (EL >= 8) {
+                bf16x8 t;
+#pragma unroll
+                for (int i = 0; i < 8; ++i) t[i] = v[ch + i];
+                *(bf16x8*)&dst[r][c + ch] = t;
+            } else {
+                bf16x4v t;
+#pragma unroll
+                for (int i = 0; i < 4; ++i) t[i] = v[ch + i];
+                *(bf16x4v*)&dst[r][c + ch] = t;
+            }
+        }
+    }
+};
 
 template <int BM, int BN, int WAVES_M, int WAVES_N,
           bool HAS_BIAS, bool RELU, bool HAS_MASK>
@@ -107,8 +135,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     constexpr int FN = WN / 16;
     constexpr int LDA = BK + LDS_PAD;
 
-    __shared__ ushort As[BM][LDA];
-    __shared__ ushort Bs[BN][LDA];
+    __shared__ ushort As[2][BM][LDA];
+    __shared__ ushort Bs[2][BN][LDA];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -127,27 +155,47 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const int lrow = lane & 15;
     const int kch = lane >> 4;  // 0..3 -> k offset kch*8
 
-    for (int k0 = 0; k0 < K; k0 += BK) {
-        // ---- stage A tile [BM][BK] and B tile [BN][BK] ----
-        stage_nt_tile<BM, HAS_MASK>(As, A, mask, M, K, m0, k0, tid);
-        stage_nt_tile<BN, false>(Bs, B, nullptr, N, K, n0, k0, tid);
-        __syncthreads();
+    StageReg<BM, HAS_MASK> ra;
+    StageReg<BN, false> rb;
+    ra.load(A, mask, M, K, m0, 0, tid);
+    rb.load(B, nullptr, N, K, n0, 0, tid);
+    ra.write(As[0], tid);
+    rb.write(Bs[0], tid);
+    __syncthreads();
 
-        // ---- MFMA ----
+    const int nsteps = (K + BK - 1) / BK;
+    int cur = 0;
+    for (int t = 0; t < nsteps; ++t) {
+        // issue next tile's global loads early (latency hides under
+        // this tile's MFMA)
+        if (t + 1 < nsteps) {
+            ra.load(A, mask, M, K, m0, (t + 1) * BK, tid);
+            rb.load(B, nullptr, N, K, n0, (t + 1) * BK, tid);
+        }
+
+        // ---- MFMA on the current tile ----
         bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
         for (int i = 0; i < FM; ++i)
-            a_frag[i] = *(const bf16x8*)&As[wm * WM + i * 16 + lrow][kch * 8];
+            a_frag[i] =
+                *(const bf16x8*)&As[cur][wm * WM + i * 16 + lrow][kch * 8];
 #pragma unroll
         for (int j = 0; j < FN; ++j)
-            b_frag[j] = *(const bf16x8*)&Bs[wn * WN + j * 16 + lrow][kch * 8];
+            b_frag[j] =
+                *(const bf16x8*)&Bs[cur][wn * WN + j * 16 + lrow][kch * 8];
 #pragma unroll
         for (int i = 0; i < FM; ++i)
 #pragma unroll
             for (int j = 0; j < FN; ++j)
                 acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+
+        if (t + 1 < nsteps) {
+            ra.write(As[cur ^ 1], tid);
+            rb.write(Bs[cur ^ 1], tid);
+        }
         __syncthreads();
+        cur ^= 1;
     }
 
     // ---- epilogue: bias, ReLU, bf16 store ----
@@ -196,8 +244,8 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     constexpr int BKW = 64;       // K-step (batch rows per stage)
     constexpr int NMB = BM / 16;  // m-blocks per tile
 
-    __shared__ ushort At[BKW / 4 * NMB * 64];  // dY^T tile, blocked
-    __shared__ ushort Bt[BKW / 4 * NMB * 64];  // X^T tile, blocked
+    __shared__ ushort At[2][BKW / 4 * NMB * 64];  // dY^T tile, blocked
+    __shared__ ushort Bt[2][BKW / 4 * NMB * 64];  // X^T tile, blocked
     __shared__ float dbs[4][64];
 
     const int tid = threadIdx.x;
@@ -228,9 +276,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const int sk = tid >> 2;
     const int sm = (tid & 3) * 16;
 
-    for (int k0 = kbeg; k0 < kend; k0 += BKW) {
+    __bf16 rva[16], rvb[16];
+    auto load_regs = [&](int k0) {
         const int gk = k0 + sk;
-        // ---- dY -> At ----
         if (gk < kend && m0 + BM <= Mo) {
 #pragma unroll
             for (int ch = 0; ch < 16; ch += 8) {
@@ -242,7 +290,8 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                     for (int i = 0; i < 8; ++i)
                         if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
                 }
-                *(bf16x8*)&At[baddr(sk, sm + ch)] = v;
+#pragma unroll
+                for (int i = 0; i < 8; ++i) rva[ch + i] = v[i];
             }
         } else {
 #pragma unroll
@@ -256,15 +305,15 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                             v = (__bf16)0.f;
                     }
                 }
-                At[baddr(sk, sm + i)] = *(const ushort*)&v;
+                rva[i] = v;
             }
         }
-        // ---- X -> Bt ----
         if (gk < kend && n0 + BN <= N) {
 #pragma unroll
             for (int ch = 0; ch < 16; ch += 8) {
                 bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + sm + ch];
-                *(bf16x8*)&Bt[baddr(sk, sm + ch)] = v;
+#pragma unroll
+                for (int i = 0; i < 8; ++i) rvb[ch + i] = v[i];
             }
         } else {
 #pragma unroll
@@ -272,15 +321,38 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                 __bf16 v = (__bf16)0.f;
                 const int gn = n0 + sm + i;
                 if (gk < kend && gn < N) v = X[(long)gk * N + gn];
-                Bt[baddr(sk, sm + i)] = *(const ushort*)&v;
+                rvb[i] = v;
             }
         }
-        __syncthreads();
+    };
+    auto write_regs = [&](int buf) {
+#pragma unroll
+        for (int ch = 0; ch < 16; ch += 8) {
+            bf16x8 va, vb;
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                va[i] = rva[ch + i];
+                vb[i] = rvb[ch + i];
+            }
+            *(bf16x8*)&At[buf][baddr(sk, sm + ch)] = va;
+            *(bf16x8*)&Bt[buf][baddr(sk, sm + ch)] = vb;
+        }
+    };
+
+    load_regs(kbeg);
+    write_regs(0);
+    __syncthreads();
+
+    const int nsteps = (kend - kbeg + BKW - 1) / BKW;
+    int cur = 0;
+    for (int t = 0; t < nsteps; ++t) {
+        if (t + 1 < nsteps) load_regs(kbeg + (t + 1) * BKW);
+
         if (do_db) {
             const int m = tid & 63;
 #pragma unroll
             for (int j = 0; j < 16; ++j) {
-                const ushort u = At[baddr((tid >> 6) * 16 + j, m)];
+                const ushort u = At[cur][baddr((tid >> 6) * 16 + j, m)];
                 db_part += bf2f(*(const __bf16*)&u);
             }
         }
@@ -296,9 +368,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int i = 0; i < 2; ++i) {
                 const int mblk = wm * 2 + i;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[(g2 * NMB + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[cur][(g2 * NMB + mblk) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[((g2 + 1) * NMB + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[cur][((g2 + 1) * NMB + mblk) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     a_frag[i][e] = lo[e];
@@ -309,9 +381,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int j = 0; j < 2; ++j) {
                 const int nblk = wn * 2 + j;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[cur][(g2 * NMB + nblk) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[cur][((g2 + 1) * NMB + nblk) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     b_frag[j][e] = lo[e];
@@ -325,7 +397,10 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
         }
+
+        if (t + 1 < nsteps) write_regs(cur ^ 1);
         __syncthreads();
+        cur ^= 1;
     }
 
     const int lrow = lane & 15;

@@ -1,6 +1,7 @@
 from .layers import (  # noqa: F401
     MLP,
     Linear,
+    MSELoss,
     Module,
     Parameter,
     ReLU,

@@ -203,6 +203,32 @@ class Softmax(Module):
         return F.softmax_bwd(dout, s)
 
 
+class MSELoss(Module):
+    """Standalone MSE loss head for API parity with the reference
+    (layers.py:145-166): forward is the IDENTITY (the loss value is
+    never computed in training, layers.py:150-155); backward takes the
+    TARGET as its dout argument and emits −2(t−x)/global_batch
+    (functional.py:43-44).  The fused SoftmaxMSE head below is the hot
+    path; this module exists for hand-built Sequential stacks.
+
+    GPU note: −2(t−x)/GB == (x−t)/(GB/2), so the backward reuses the
+    fused xent elementwise kernel with a halved batch scale."""
+
+    def __init__(self, global_batch_size: int):
+        super().__init__()
+        self.global_batch_size = global_batch_size
+
+    def forward(self, inputs, mubatch_id: int = 0):
+        self._stash("x", mubatch_id, inputs)
+        return inputs
+
+    def backward(self, target, mubatch_id: int = 0):
+        x = self._unstash("x", mubatch_id)
+        if target.dtype != x.dtype:
+            target = target.to(x.dtype)
+        return F.head_softmax_xent_bwd(x, target, self.global_batch_size / 2.0)
+
+
 class _LossHead(Module):
     """Fused softmax+loss head.
 

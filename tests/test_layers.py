@@ -150,3 +150,28 @@ def test_eval_mode_no_stash():
     model.eval()
     model.forward(torch.randn(2, 4), 0)
     assert len(model.layers[0]._cache) == 0
+
+
+def test_standalone_mse_loss_module():
+    """Softmax + MSELoss modules chained == fused SoftmaxMSE head
+    (reference layers.py:83-96 + 145-166 as separate modules)."""
+    from shallowspeed_amd.models import MSELoss, Softmax
+
+    gb = 8
+    x = torch.randn(8, 6)
+    t = torch.zeros(8, 6)
+    t[torch.arange(8), torch.randint(0, 6, (8,))] = 1
+
+    chain = Sequential([Linear(10, 6), Softmax(), MSELoss(gb)])
+    fused = Sequential([Linear(10, 6), SoftmaxMSE(gb)])
+    xin = torch.randn(8, 10)
+    y1 = chain.forward(xin, 0)
+    y2 = fused.forward(xin, 0)
+    torch.testing.assert_close(y1, y2)
+    d1 = chain.backward(t, 0)
+    d2 = fused.backward(t, 0)
+    torch.testing.assert_close(d1, d2)
+    # loss value helper (reference functional.py:38-40)
+    from shallowspeed_amd.ops import functional as F
+
+    assert F.mse_loss(y1, t, gb).item() >= 0

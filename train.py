@@ -79,6 +79,8 @@ def main():
     ap.add_argument("--samples", type=int, default=8192)
     ap.add_argument("--device", default=None, help="cpu | cuda")
     ap.add_argument("--save", default=None, help="checkpoint dir to write")
+    ap.add_argument("--timing", action="store_true",
+                    help="print per-instruction wall time after training")
     ap.add_argument("--resume", default=None, help="checkpoint dir to read")
     args = ap.parse_args()
 
@@ -110,6 +112,8 @@ def main():
     val_ds.load(topo.dp_rank, args.dp)
 
     worker = Worker(topo, model, train_ds, optimizer)
+    if args.timing:
+        worker.enable_instruction_timing(True)
     val_worker = Worker(topo, model, val_ds, None, use_dp=False)
     sched_cls = SCHEDULES[args.schedule]
 
@@ -137,6 +141,13 @@ def main():
         from shallowspeed_amd.checkpoint import save_checkpoint
 
         save_checkpoint(args.save, model, topo, step=args.epochs)
+
+    if args.timing and topo.rank == 0:
+        total = sum(worker.instruction_times.values())
+        print("instruction time breakdown:")
+        for k, v in sorted(worker.instruction_times.items(),
+                           key=lambda kv: -kv[1]):
+            print(f"  {k:24s} {v:8.3f}s ({v/total*100:5.1f}%)")
 
     # replica-sync invariant (reference train.py:154-155)
     if topo.dp > 1:

@@ -222,7 +222,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
 // ---------------------------------------------------------------- wgrad
 
-template <bool HAS_MASK>
+template <int BMT, int BNT, bool HAS_MASK>
 __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const __bf16* __restrict__ dY,    // [Kb][Mo]
     const __bf16* __restrict__ X,     // [Kb][N]
@@ -231,7 +231,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     float* __restrict__ gb,           // [Mo] or null: fused bias grad
     int Mo, int N, int Kb, int k_per_split) {
     // TN GEMM over the batch axis.  Tiles are staged in a BLOCKED
-    // [BK/4][BM/16][4][16] bf16 layout so that
+    // [BKW/4][rows/16][4][16] bf16 layout so that
     //   * staging is plain 16-B vector writes (global rows are
     //     m/n-contiguous — no software transpose), and
     //   * MFMA A/B fragments come out of LDS through the gfx950
@@ -240,119 +240,112 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     //     group hands the crossbar one [4][16] subtile and receives
     //     its column — semantics verified on hardware by
     //     scripts/probe_tr.hip.
-    constexpr int BM = 64, BN = 64;
-    constexpr int BKW = 64;       // K-step (batch rows per stage)
-    constexpr int NMB = BM / 16;  // m-blocks per tile
+    // The 128x128 instantiation halves both operands' L2 re-reads for
+    // wide layers (traffic ∝ dY·(N/BNT) + X·(Mo/BMT)).
+    constexpr int BKW = 64;        // K-step (batch rows per stage)
+    constexpr int NMA = BMT / 16;  // 16-row blocks per A tile
+    constexpr int NMB = BNT / 16;
+    constexpr int FM = BMT / 32;   // 16x16 frags per wave (2x2 waves)
+    constexpr int FN = BNT / 32;
 
-    __shared__ ushort At[2][BKW / 4 * NMB * 64];  // dY^T tile, blocked
-    __shared__ ushort Bt[2][BKW / 4 * NMB * 64];  // X^T tile, blocked
-    __shared__ float dbs[4][64];
+    __shared__ ushort At[BKW / 4 * NMA * 64];  // dY^T tile, blocked
+    __shared__ ushort Bt[BKW / 4 * NMB * 64];  // X^T tile, blocked
+    __shared__ float dbs[256 / BMT > 1 ? 256 / BMT : 2][BMT];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
     const int wm = wave >> 1;  // 2x2 waves
     const int wn = wave & 1;
-    const int m0 = blockIdx.x * BM;
-    const int n0 = blockIdx.y * BN;
+    const int m0 = blockIdx.x * BMT;
+    const int n0 = blockIdx.y * BNT;
     const int kbeg = blockIdx.z * k_per_split;
     const int kend = min(Kb, kbeg + k_per_split);
 
-    f32x4 acc[2][2];
+    f32x4 acc[FM][FN];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < FM; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     const bool do_db = (gb != nullptr) && (blockIdx.y == 0);
     float db_part = 0.f;
 
-    // blocked-layout element address
-    auto baddr = [](int k, int m) {
+    auto baddrA = [](int k, int m) {
+        return ((k >> 2) * NMA + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
+    };
+    auto baddrB = [](int k, int m) {
         return ((k >> 2) * NMB + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
     };
 
-    // staging geometry: thread t covers (k = t/4, m = 16*(t%4) .. +16)
-    const int sk = tid >> 2;
-    const int sm = (tid & 3) * 16;
+    // staging: thread t covers ELA contiguous elems of the [BKW][BMT]
+    // tile (vector fast path when fully in range)
+    constexpr int ELA = BKW * BMT / 256;
+    constexpr int ELB = BKW * BNT / 256;
+    const int ska = (tid * ELA) / BMT, sma = (tid * ELA) % BMT;
+    const int skb = (tid * ELB) / BNT, smb = (tid * ELB) % BNT;
 
-    __bf16 rva[16], rvb[16];
-    auto load_regs = [&](int k0) {
-        const int gk = k0 + sk;
-        if (gk < kend && m0 + BM <= Mo) {
+    for (int k0 = kbeg; k0 < kend; k0 += BKW) {
+        {
+            const int gk = k0 + ska;
+            if (gk < kend && m0 + BMT <= Mo) {
 #pragma unroll
-            for (int ch = 0; ch < 16; ch += 8) {
-                bf16x8 v = *(const bf16x8*)&dY[(long)gk * Mo + m0 + sm + ch];
-                if constexpr (HAS_MASK) {
-                    bf16x8 mk =
-                        *(const bf16x8*)&mask[(long)gk * Mo + m0 + sm + ch];
-#pragma unroll
-                    for (int i = 0; i < 8; ++i)
-                        if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
-                }
-#pragma unroll
-                for (int i = 0; i < 8; ++i) rva[ch + i] = v[i];
-            }
-        } else {
-#pragma unroll
-            for (int i = 0; i < 16; ++i) {
-                __bf16 v = (__bf16)0.f;
-                const int gm = m0 + sm + i;
-                if (gk < kend && gm < Mo) {
-                    v = dY[(long)gk * Mo + gm];
+                for (int ch = 0; ch < ELA; ch += 8) {
+                    bf16x8 v =
+                        *(const bf16x8*)&dY[(long)gk * Mo + m0 + sma + ch];
                     if constexpr (HAS_MASK) {
-                        if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
-                            v = (__bf16)0.f;
+                        bf16x8 mk = *(const bf16x8*)&mask[(long)gk * Mo + m0 +
+                                                          sma + ch];
+#pragma unroll
+                        for (int i = 0; i < 8; ++i)
+                            if (!(bf2f(mk[i]) > 0.f)) v[i] = (__bf16)0.f;
                     }
+                    *(bf16x8*)&At[baddrA(ska, sma + ch)] = v;
                 }
-                rva[i] = v;
+            } else {
+#pragma unroll
+                for (int i = 0; i < ELA; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    const int gm = m0 + sma + i;
+                    if (gk < kend && gm < Mo) {
+                        v = dY[(long)gk * Mo + gm];
+                        if constexpr (HAS_MASK) {
+                            if (!(bf2f(mask[(long)gk * Mo + gm]) > 0.f))
+                                v = (__bf16)0.f;
+                        }
+                    }
+                    At[baddrA(ska, sma + i)] = *(const ushort*)&v;
+                }
             }
         }
-        if (gk < kend && n0 + BN <= N) {
+        {
+            const int gk = k0 + skb;
+            if (gk < kend && n0 + BNT <= N) {
 #pragma unroll
-            for (int ch = 0; ch < 16; ch += 8) {
-                bf16x8 v = *(const bf16x8*)&X[(long)gk * N + n0 + sm + ch];
+                for (int ch = 0; ch < ELB; ch += 8) {
+                    bf16x8 v =
+                        *(const bf16x8*)&X[(long)gk * N + n0 + smb + ch];
+                    *(bf16x8*)&Bt[baddrB(skb, smb + ch)] = v;
+                }
+            } else {
 #pragma unroll
-                for (int i = 0; i < 8; ++i) rvb[ch + i] = v[i];
-            }
-        } else {
-#pragma unroll
-            for (int i = 0; i < 16; ++i) {
-                __bf16 v = (__bf16)0.f;
-                const int gn = n0 + sm + i;
-                if (gk < kend && gn < N) v = X[(long)gk * N + gn];
-                rvb[i] = v;
+                for (int i = 0; i < ELB; ++i) {
+                    __bf16 v = (__bf16)0.f;
+                    const int gn = n0 + smb + i;
+                    if (gk < kend && gn < N) v = X[(long)gk * N + gn];
+                    Bt[baddrB(skb, smb + i)] = *(const ushort*)&v;
+                }
             }
         }
-    };
-    auto write_regs = [&](int buf) {
-#pragma unroll
-        for (int ch = 0; ch < 16; ch += 8) {
-            bf16x8 va, vb;
-#pragma unroll
-            for (int i = 0; i < 8; ++i) {
-                va[i] = rva[ch + i];
-                vb[i] = rvb[ch + i];
-            }
-            *(bf16x8*)&At[buf][baddr(sk, sm + ch)] = va;
-            *(bf16x8*)&Bt[buf][baddr(sk, sm + ch)] = vb;
-        }
-    };
-
-    load_regs(kbeg);
-    write_regs(0);
-    __syncthreads();
-
-    const int nsteps = (kend - kbeg + BKW - 1) / BKW;
-    int cur = 0;
-    for (int t = 0; t < nsteps; ++t) {
-        if (t + 1 < nsteps) load_regs(kbeg + (t + 1) * BKW);
+        __syncthreads();
 
         if (do_db) {
-            const int m = tid & 63;
+            constexpr int NKQ = 256 / BMT;       // threads stacked on k
+            constexpr int KPT = BKW / NKQ;       // k values per thread
+            const int m = tid % BMT, kq = tid / BMT;
 #pragma unroll
-            for (int j = 0; j < 16; ++j) {
-                const ushort u = At[cur][baddr((tid >> 6) * 16 + j, m)];
+            for (int j = 0; j < KPT; ++j) {
+                const ushort u = At[baddrA(kq * KPT + j, m)];
                 db_part += bf2f(*(const __bf16*)&u);
             }
         }
@@ -363,14 +356,14 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #pragma unroll
         for (int kk = 0; kk < BKW / 32; ++kk) {
             const int g2 = kk * 8 + (lane >> 4) * 2;  // k-subtile pair
-            bf16x8 a_frag[2], b_frag[2];
+            bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
-            for (int i = 0; i < 2; ++i) {
-                const int mblk = wm * 2 + i;
+            for (int i = 0; i < FM; ++i) {
+                const int mblk = wm * FM + i;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[cur][(g2 * NMB + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[(g2 * NMA + mblk) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[cur][((g2 + 1) * NMB + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[((g2 + 1) * NMA + mblk) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     a_frag[i][e] = lo[e];
@@ -378,12 +371,12 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                 }
             }
 #pragma unroll
-            for (int j = 0; j < 2; ++j) {
-                const int nblk = wn * 2 + j;
+            for (int j = 0; j < FN; ++j) {
+                const int nblk = wn * FN + j;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[cur][(g2 * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[cur][((g2 + 1) * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     b_frag[j][e] = lo[e];
@@ -391,29 +384,26 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                 }
             }
 #pragma unroll
-            for (int i = 0; i < 2; ++i)
+            for (int i = 0; i < FM; ++i)
 #pragma unroll
-                for (int j = 0; j < 2; ++j)
+                for (int j = 0; j < FN; ++j)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
         }
-
-        if (t + 1 < nsteps) write_regs(cur ^ 1);
         __syncthreads();
-        cur ^= 1;
     }
 
     const int lrow = lane & 15;
     const int kch = lane >> 4;
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < FM; ++i) {
 #pragma unroll
-        for (int j = 0; j < 2; ++j) {
-            const int gcol = n0 + wn * 32 + j * 16 + lrow;
+        for (int j = 0; j < FN; ++j) {
+            const int gcol = n0 + wn * (BNT / 2) + j * 16 + lrow;
             if (gcol >= N) continue;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int grow = m0 + wm * 32 + i * 16 + kch * 4 + r;
+                const int grow = m0 + wm * (BMT / 2) + i * 16 + kch * 4 + r;
                 if (grow >= Mo) continue;
                 atomicAdd(&gW[(long)grow * N + gcol], acc[i][j][r]);
             }
@@ -421,11 +411,13 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     }
 
     if (do_db) {
-        dbs[tid >> 6][tid & 63] = db_part;
+        constexpr int NKQ = 256 / BMT;
+        dbs[tid / BMT][tid % BMT] = db_part;
         __syncthreads();
-        if (tid < 64 && m0 + tid < Mo) {
-            const float s4 = dbs[0][tid] + dbs[1][tid] + dbs[2][tid] +
-                             dbs[3][tid];
+        if (tid < BMT && m0 + tid < Mo) {
+            float s4 = 0.f;
+#pragma unroll
+            for (int q = 0; q < NKQ; ++q) s4 += dbs[q][tid];
             atomicAdd(&gb[m0 + tid], s4);
         }
     }
@@ -509,26 +501,29 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
 void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
                  void* gb, int Mo, int N, int Kb, int split_k,
                  hipStream_t stream) {
-    // pick split so that grid covers the CUs: tiles * split >= ~512
+    const bool big = Mo >= 128 && N >= 128;
+    const int bm = big ? 128 : 64, bn = big ? 128 : 64;
+    // pick split so that grid covers the CUs (~2-4 blocks/CU)
     if (split_k <= 0) {
-        const int tiles = cdiv(Mo, 64) * cdiv(N, 64);
+        const int tiles = cdiv(Mo, bm) * cdiv(N, bn);
         split_k = 1;
         while (tiles * split_k < 1024 && split_k * 128 < Kb) split_k *= 2;
     }
-    int k_per_split = cdiv(cdiv(Kb, split_k), 64) * 64;  // wgrad BKW=64
+    int k_per_split = cdiv(cdiv(Kb, split_k), 64) * 64;  // BKW=64
     split_k = cdiv(Kb, k_per_split);
-    dim3 grid(cdiv(Mo, 64), cdiv(N, 64), split_k);
+    dim3 grid(cdiv(Mo, bm), cdiv(N, bn), split_k);
     dim3 block(256);
-    if (mask)
-        hipLaunchKernelGGL((wgrad_tn_kernel<true>), grid, block, 0, stream,
-                           (const __bf16*)dY, (const __bf16*)X,
-                           (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,
-                           Kb, k_per_split);
-    else
-        hipLaunchKernelGGL((wgrad_tn_kernel<false>), grid, block, 0, stream,
-                           (const __bf16*)dY, (const __bf16*)X,
-                           (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,
-                           Kb, k_per_split);
+#define WLAUNCH(BMT, BNT, HM)                                               \
+    hipLaunchKernelGGL((wgrad_tn_kernel<BMT, BNT, HM>), grid, block, 0,     \
+                       stream, (const __bf16*)dY, (const __bf16*)X,         \
+                       (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,  \
+                       Kb, k_per_split)
+    if (big) {
+        if (mask) WLAUNCH(128, 128, true); else WLAUNCH(128, 128, false);
+    } else {
+        if (mask) WLAUNCH(64, 64, true); else WLAUNCH(64, 64, false);
+    }
+#undef WLAUNCH
 }
 
 void ss_colsum(const void* dY, const void* mask, void* gb, int M, int N,

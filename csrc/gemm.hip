@@ -501,7 +501,10 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
 void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
                  void* gb, int Mo, int N, int Kb, int split_k,
                  hipStream_t stream) {
-    const bool big = Mo >= 128 && N >= 128;
+    // big tiles only pay when both dims are wide enough that the
+    // grid still covers the CUs AND operand re-reads dominate
+    // (measured: at 256-wide layers the 64-config's block count wins)
+    const bool big = Mo >= 512 && N >= 512;
     const int bm = big ? 128 : 64, bn = big ? 128 : 64;
     // pick split so that grid covers the CUs (~2-4 blocks/CU)
     if (split_k <= 0) {

@@ -25,14 +25,29 @@ import torch
 
 def synthesize(n_samples: int, in_dim: int = 784, n_classes: int = 10,
                seed: int = 1234, teacher_seed: int = 99991):
-    """The TEACHER is seeded independently of the split seed so train
-    and validation splits are labeled by the same function (otherwise
-    val accuracy is stuck at chance)."""
+    """MNIST-shaped synthetic data labeled by a fixed random linear
+    teacher.
+
+    * The TEACHER is seeded independently of the split seed so train
+      and validation splits are labeled by the same function
+      (otherwise val accuracy is stuck at chance).
+    * Samples are MARGIN-FILTERED: 2× candidates are generated and the
+      half with the largest top1−top2 teacher-logit gap is kept
+      (deterministically), so the task is cleanly learnable and
+      convergence gates ("accuracy climbs", reference train.py:148-152)
+      are crisp.
+    """
     g = torch.Generator().manual_seed(seed)
-    x = torch.randn(n_samples, in_dim, generator=g, dtype=torch.float32)
+    x = torch.randn(2 * n_samples, in_dim, generator=g, dtype=torch.float32)
     gt = torch.Generator().manual_seed(teacher_seed + in_dim * 31 + n_classes)
     teacher = torch.randn(in_dim, n_classes, generator=gt, dtype=torch.float32)
-    labels = (x @ teacher).argmax(dim=1)
+    logits = x @ teacher
+    top2 = logits.topk(min(2, n_classes), dim=1).values
+    margin = top2[:, 0] - (top2[:, 1] if n_classes > 1 else 0)
+    keep = margin.argsort(descending=True)[:n_samples]
+    keep = keep.sort().values  # preserve original order
+    x = x[keep].contiguous()
+    labels = logits[keep].argmax(dim=1)
     y = torch.zeros(n_samples, n_classes, dtype=torch.float32)
     y[torch.arange(n_samples), labels] = 1.0
     return x, y

@@ -54,6 +54,8 @@ def main():
     is_gpu = device.type == "cuda"
 
     sizes = [int(s) for s in args.layer_sizes.split(",")]
+    assert args.local_batch % args.mubatches == 0
+    assert len(sizes) % pp == 0, "layer count must divide into pp stages"
     global_batch = args.local_batch * dp
     mubatch = args.local_batch // args.mubatches
 

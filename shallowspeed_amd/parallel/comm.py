@@ -86,6 +86,9 @@ def init_topology(dp: int, pp: int, backend: Optional[str] = None,
         else:
             device = torch.device("cpu")
     if device.type == "cuda":
+        if device.index is None:
+            local = int(os.environ.get("LOCAL_RANK", rank))
+            device = torch.device("cuda", local % torch.cuda.device_count())
         torch.cuda.set_device(device)
 
     if world == 1:

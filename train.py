@@ -84,6 +84,10 @@ def main():
     ap.add_argument("--resume", default=None, help="checkpoint dir to read")
     args = ap.parse_args()
 
+    assert args.global_batch % (args.dp * args.mubatches) == 0, \
+        "--global-batch must divide by dp*mubatches"
+    assert len(args.layer_sizes) % args.pp == 0, \
+        "len(--layer-sizes) must divide into --pp stages (layers.py:242)"
     device = torch.device(args.device) if args.device else None
     topo = init_topology(args.dp, args.pp, device=device)
     device = topo.device

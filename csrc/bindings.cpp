@@ -24,6 +24,14 @@ void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
 void ss_head_xent_bwd(const void*, const void*, void*, long, float,
                       hipStream_t);
 void ss_sgd_multi(const void*, int, long, float, hipStream_t);
+void ss_ln_fwd(const void*, const void*, const void*, void*, void*, void*,
+               int, int, float, hipStream_t);
+void ss_ln_bwd_dx(const void*, const void*, const void*, const void*,
+                  const void*, void*, int, int, hipStream_t);
+void ss_ln_bwd_dparam(const void*, const void*, const void*, const void*,
+                      void*, void*, int, int, hipStream_t);
+void ss_gelu_fwd(const void*, void*, long, hipStream_t);
+void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
 }
 
 namespace {
@@ -170,6 +178,66 @@ torch::Tensor head_xent_bwd(torch::Tensor s, torch::Tensor t,
     return dz;
 }
 
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta, double eps) {
+    check_bf16(x, "x");
+    check_bf16(gamma, "gamma");
+    check_bf16(beta, "beta");
+    TORCH_CHECK(x.dim() == 2, "x must be 2-D");
+    const int B = x.size(0), C = x.size(1);
+    TORCH_CHECK(gamma.numel() == C && beta.numel() == C, "param size");
+    auto y = torch::empty_like(x);
+    auto mean = torch::empty({B}, x.options().dtype(torch::kFloat));
+    auto rstd = torch::empty({B}, x.options().dtype(torch::kFloat));
+    ss_ln_fwd(x.data_ptr(), gamma.data_ptr(), beta.data_ptr(), y.data_ptr(),
+              mean.data_ptr(), rstd.data_ptr(), B, C, (float)eps,
+              cur_stream());
+    return {y, mean, rstd};
+}
+
+torch::Tensor ln_bwd_dx(torch::Tensor dy, torch::Tensor x,
+                        torch::Tensor gamma, torch::Tensor mean,
+                        torch::Tensor rstd) {
+    check_bf16(dy, "dy");
+    check_bf16(x, "x");
+    check_bf16(gamma, "gamma");
+    check_f32(mean, "mean");
+    check_f32(rstd, "rstd");
+    auto dx = torch::empty_like(dy);
+    ss_ln_bwd_dx(dy.data_ptr(), x.data_ptr(), gamma.data_ptr(),
+                 mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(), dy.size(0),
+                 dy.size(1), cur_stream());
+    return dx;
+}
+
+void ln_bwd_dparam(torch::Tensor dy, torch::Tensor x, torch::Tensor mean,
+                   torch::Tensor rstd, torch::Tensor dgamma,
+                   torch::Tensor dbeta) {
+    check_bf16(dy, "dy");
+    check_bf16(x, "x");
+    check_f32(dgamma, "dgamma");
+    check_f32(dbeta, "dbeta");
+    ss_ln_bwd_dparam(dy.data_ptr(), x.data_ptr(), mean.data_ptr(),
+                     rstd.data_ptr(), dgamma.data_ptr(), dbeta.data_ptr(),
+                     dy.size(0), dy.size(1), cur_stream());
+}
+
+torch::Tensor gelu_fwd(torch::Tensor z) {
+    check_bf16(z, "z");
+    auto y = torch::empty_like(z);
+    ss_gelu_fwd(z.data_ptr(), y.data_ptr(), z.numel(), cur_stream());
+    return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor z) {
+    check_bf16(dy, "dy");
+    check_bf16(z, "z");
+    auto dz = torch::empty_like(dy);
+    ss_gelu_bwd(dy.data_ptr(), z.data_ptr(), dz.data_ptr(), dy.numel(),
+                cur_stream());
+    return dz;
+}
+
 void sgd_multi(torch::Tensor desc, double lr, int64_t total) {
     // total passed by the caller (cached host-side) — no device sync.
     TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
@@ -193,4 +261,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_mse_bwd", &head_mse_bwd);
     m.def("head_xent_bwd", &head_xent_bwd);
     m.def("sgd_multi", &sgd_multi);
+    m.def("ln_fwd", &ln_fwd);
+    m.def("ln_bwd_dx", &ln_bwd_dx);
+    m.def("ln_bwd_dparam", &ln_bwd_dparam);
+    m.def("gelu_fwd", &gelu_fwd);
+    m.def("gelu_bwd", &gelu_bwd);
 }

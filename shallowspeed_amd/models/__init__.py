@@ -1,5 +1,7 @@
 from .layers import (  # noqa: F401
+    GELU,
     MLP,
+    LayerNorm,
     Linear,
     MSELoss,
     Module,

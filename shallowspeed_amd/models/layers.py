@@ -139,7 +139,7 @@ class Linear(Module):
 
     def __init__(self, in_dims: int, out_dims: int, activation=None):
         super().__init__()
-        assert activation in (None, "relu")
+        assert activation in (None, "relu", "gelu")
         self.in_dims, self.out_dims = in_dims, out_dims
         self.activation = activation
         g = torch.Generator().manual_seed(_shape_seed(in_dims, out_dims))
@@ -152,6 +152,11 @@ class Linear(Module):
 
     def forward(self, inputs, mubatch_id: int = 0):
         self._stash("x", mubatch_id, inputs)
+        if self.activation == "gelu":
+            z = F.linear_fwd(inputs, self.weight.compute(),
+                             self.bias.compute(), relu=False)
+            self._stash("z", mubatch_id, z)
+            return F.gelu_fwd(z)
         y = F.linear_fwd(
             inputs, self.weight.compute(), self.bias.compute(),
             relu=(self.activation == "relu"),
@@ -165,7 +170,12 @@ class Linear(Module):
 
     def backward(self, dout, mubatch_id: int = 0, need_dx: bool = True):
         x = self._unstash("x", mubatch_id)
-        mask_src = self._unstash("y", mubatch_id) if self.activation == "relu" else None
+        mask_src = None
+        if self.activation == "relu":
+            mask_src = self._unstash("y", mubatch_id)
+        elif self.activation == "gelu":
+            z = self._unstash("z", mubatch_id)
+            dout = F.gelu_bwd(dout, z)
         dx = None
         if need_dx:
             dx = F.linear_dgrad(dout, self.weight.compute(),
@@ -186,6 +196,45 @@ class ReLU(Module):
     def backward(self, dout, mubatch_id: int = 0):
         y = self._unstash("y", mubatch_id)
         return F.relu_bwd(dout, y)
+
+
+class GELU(Module):
+    """Standalone GELU (tanh approximation) — beyond-reference module
+    for modern MLP blocks; the fused path is Linear(activation="gelu")."""
+
+    def forward(self, inputs, mubatch_id: int = 0):
+        self._stash("z", mubatch_id, inputs)
+        return F.gelu_fwd(inputs)
+
+    def backward(self, dout, mubatch_id: int = 0):
+        z = self._unstash("z", mubatch_id)
+        return F.gelu_bwd(dout, z)
+
+
+class LayerNorm(Module):
+    """Rowwise LayerNorm with learnable scale/shift — beyond-reference
+    module (HIP kernels: csrc/norm.hip; wave-per-row f32 statistics)."""
+
+    def __init__(self, dim: int, eps: float = 1e-5):
+        super().__init__()
+        self.dim, self.eps = dim, eps
+        self._params["gamma"] = Parameter(torch.ones(dim))
+        self._params["beta"] = Parameter(torch.zeros(dim))
+        self.gamma = self._params["gamma"]
+        self.beta = self._params["beta"]
+
+    def forward(self, inputs, mubatch_id: int = 0):
+        y, mean, rstd = F.layernorm_fwd(
+            inputs, self.gamma.compute(), self.beta.compute(), self.eps)
+        self._stash("x", mubatch_id, inputs)
+        self._stash("stats", mubatch_id, (mean, rstd))
+        return y
+
+    def backward(self, dout, mubatch_id: int = 0):
+        x = self._unstash("x", mubatch_id)
+        mean, rstd = self._unstash("stats", mubatch_id)
+        return F.layernorm_bwd(dout, x, self.gamma.compute(), mean, rstd,
+                               self.gamma.grad, self.beta.grad)
 
 
 class Softmax(Module):

@@ -182,3 +182,53 @@ def head_softmax_xent_bwd(probs, target, global_batch):
 def xent_loss(probs, target, global_batch, eps=1e-9):
     """−Σ t·log(p)/GB (logging only)."""
     return -(target * torch.log(probs.float() + eps)).sum() / global_batch
+
+
+# ----------------------------------------------- layernorm / gelu (extras)
+
+def layernorm_fwd(x, gamma, beta, eps=1e-5):
+    """Rowwise LayerNorm.  Returns (y, mean, rstd) — the f32 row stats
+    feed the backward.  Beyond-reference extension (modern-MLP blocks);
+    HIP kernel: csrc/norm.hip."""
+    if _is_gpu(x):
+        return tuple(_ext_for(x).ln_fwd(x, gamma, beta, float(eps)))
+    xf = x.float()
+    mu = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    rstd = (var + eps).rsqrt()
+    y = ((xf - mu) * rstd * gamma.float() + beta.float()).to(x.dtype)
+    return y, mu.squeeze(-1), rstd.squeeze(-1)
+
+
+def layernorm_bwd(dy, x, gamma, mean, rstd, dgamma, dbeta):
+    """dx plus in-place accumulation of dγ/dβ (f32 grads)."""
+    if _is_gpu(dy):
+        e = _ext_for(dy)
+        dx = e.ln_bwd_dx(dy, x, gamma, mean, rstd)
+        e.ln_bwd_dparam(dy, x, mean, rstd, dgamma, dbeta)
+        return dx
+    xf, dyf = x.float(), dy.float()
+    xh = (xf - mean.unsqueeze(-1)) * rstd.unsqueeze(-1)
+    g = dyf * gamma.float()
+    C = x.shape[-1]
+    dx = rstd.unsqueeze(-1) * (
+        g - g.mean(-1, keepdim=True) - xh * (g * xh).mean(-1, keepdim=True))
+    dgamma += (dyf * xh).sum(0)
+    dbeta += dyf.sum(0)
+    return dx.to(dy.dtype)
+
+
+def gelu_fwd(z):
+    """tanh-approximation GELU (elementwise HIP kernel on GPU)."""
+    if _is_gpu(z):
+        return _ext_for(z).gelu_fwd(z)
+    return torch.nn.functional.gelu(z.float(), approximate="tanh").to(z.dtype)
+
+
+def gelu_bwd(dy, z):
+    if _is_gpu(dy):
+        return _ext_for(dy).gelu_bwd(dy, z)
+    zf = z.float().detach().requires_grad_(True)
+    y = torch.nn.functional.gelu(zf, approximate="tanh")
+    y.backward(dy.float())
+    return zf.grad.to(dy.dtype)

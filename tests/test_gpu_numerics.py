@@ -306,3 +306,81 @@ def test_gpu_checkpoint_roundtrip(tmp_path, gpu_device):
         torch.testing.assert_close(a.lp.float(), b.lp.float())
         if a.lp_t is not None:
             torch.testing.assert_close(a.lp_t.float(), b.lp_t.float())
+
+
+# --------------------------------------------------- layernorm / gelu
+
+@pytest.mark.parametrize("C", [32, 256, 777])
+def test_gpu_layernorm_vs_torch(gpu_device, C):
+    e = ext()
+    B = 513
+    x = rand_bf16(B, C, device=gpu_device, seed=21)
+    gamma = rand_bf16(C, device=gpu_device, seed=22)
+    beta = rand_bf16(C, device=gpu_device, seed=23)
+    y, mean, rstd = e.ln_fwd(x, gamma, beta, 1e-5)
+    ref = torch.nn.functional.layer_norm(
+        x.float(), (C,), gamma.float(), beta.float(), 1e-5)
+    torch.testing.assert_close(y.float(), ref, atol=3e-2, rtol=3e-2)
+
+    dy = rand_bf16(B, C, device=gpu_device, seed=24)
+    dx = e.ln_bwd_dx(dy, x, gamma, mean, rstd)
+    dgamma = torch.zeros(C, device=gpu_device, dtype=torch.float32)
+    dbeta = torch.zeros(C, device=gpu_device, dtype=torch.float32)
+    e.ln_bwd_dparam(dy, x, mean, rstd, dgamma, dbeta)
+
+    xr = x.float().requires_grad_(True)
+    gr = gamma.float().requires_grad_(True)
+    br = beta.float().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(xr, (C,), gr, br, 1e-5)
+    yr.backward(dy.float())
+    torch.testing.assert_close(dx.float(), xr.grad, atol=3e-2, rtol=5e-2)
+    torch.testing.assert_close(dgamma, gr.grad, atol=0.3, rtol=3e-2)
+    torch.testing.assert_close(dbeta, br.grad, atol=0.3, rtol=3e-2)
+
+
+def test_gpu_gelu_vs_torch(gpu_device):
+    e = ext()
+    z = rand_bf16(1000, 53, device=gpu_device, scale=2.0, seed=25)
+    y = e.gelu_fwd(z)
+    ref = torch.nn.functional.gelu(z.float(), approximate="tanh")
+    torch.testing.assert_close(y.float(), ref, atol=1e-2, rtol=2e-2)
+    dy = rand_bf16(1000, 53, device=gpu_device, seed=26)
+    zf = z.float().requires_grad_(True)
+    torch.nn.functional.gelu(zf, approximate="tanh").backward(dy.float())
+    dz = e.gelu_bwd(dy, z)
+    torch.testing.assert_close(dz.float(), zf.grad, atol=1e-2, rtol=2e-2)
+
+
+def test_gpu_ln_gelu_stack_trains(gpu_device):
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import (LayerNorm, Linear, SGD, Sequential,
+                                         SoftmaxXent)
+    from shallowspeed_amd.parallel import NaiveParallelSchedule, Topology, Worker
+
+    gbs = 256
+    model = Sequential([
+        Linear(64, 128, activation="gelu"),
+        LayerNorm(128),
+        Linear(128, 10),
+        SoftmaxXent(gbs),
+    ])
+    model.in_dim, model.out_dim = 64, 10
+    model.materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.1)
+    ds = Dataset(gbs, 64, n_samples=1024, in_dim=64, n_classes=10,
+                 device=gpu_device).load(0, 1)
+    w = Worker(Topology(device=gpu_device), model, ds, opt)
+
+    def acc():
+        model.eval()
+        p = model.forward(ds.x.to(BF), 0)
+        model.train()
+        return (p.float().argmax(-1) == ds.y.float().argmax(-1)).float() \
+            .mean().item()
+
+    a0 = acc()
+    for _ in range(20):
+        for b in range(ds.num_batches()):
+            w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
+    torch.cuda.synchronize()
+    assert acc() > a0 + 0.15

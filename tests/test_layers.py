@@ -175,3 +175,82 @@ def test_standalone_mse_loss_module():
     from shallowspeed_amd.ops import functional as F
 
     assert F.mse_loss(y1, t, gb).item() >= 0
+
+
+def test_layernorm_gelu_vs_autograd():
+    """LayerNorm + GELU modules (beyond-reference extensions) against
+    torch.autograd."""
+    from shallowspeed_amd.models import GELU, LayerNorm
+
+    torch.manual_seed(3)
+    B, C = 16, 32
+    model = Sequential([LayerNorm(C), GELU()])
+    x = torch.randn(B, C)
+    dy = torch.randn(B, C)
+    y = model.forward(x, 0)
+    dx = model.backward(dy, 0)
+
+    xr = x.clone().requires_grad_(True)
+    g = torch.ones(C, requires_grad=True)
+    b = torch.zeros(C, requires_grad=True)
+    yr = torch.nn.functional.gelu(
+        torch.nn.functional.layer_norm(xr, (C,), g, b), approximate="tanh")
+    yr.backward(dy)
+    torch.testing.assert_close(y, yr.detach(), atol=2e-5, rtol=2e-5)
+    torch.testing.assert_close(dx, xr.grad, atol=2e-5, rtol=2e-5)
+    ln = model.layers[0]
+    torch.testing.assert_close(ln.gamma.grad, g.grad, atol=2e-5, rtol=2e-5)
+    torch.testing.assert_close(ln.beta.grad, b.grad, atol=2e-5, rtol=2e-5)
+
+
+def test_linear_gelu_fused_vs_autograd():
+    torch.manual_seed(4)
+    model = Sequential([Linear(12, 8, activation="gelu")])
+    x = torch.randn(6, 12)
+    dy = torch.randn(6, 8)
+    y = model.forward(x, 0)
+    dx = model.backward(dy, 0)
+
+    w = model.layers[0].weight.data.clone().requires_grad_(True)
+    b = model.layers[0].bias.data.clone().requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+    yr = torch.nn.functional.gelu(xr @ w.t() + b, approximate="tanh")
+    yr.backward(dy)
+    torch.testing.assert_close(y, yr.detach(), atol=2e-5, rtol=2e-5)
+    torch.testing.assert_close(dx, xr.grad, atol=2e-5, rtol=2e-5)
+    torch.testing.assert_close(model.layers[0].weight.grad, w.grad,
+                               atol=2e-5, rtol=2e-5)
+    torch.testing.assert_close(model.layers[0].bias.grad, b.grad,
+                               atol=2e-5, rtol=2e-5)
+
+
+def test_layernorm_mlp_trains():
+    """A LN+GELU MLP block stack trains end-to-end through the Worker."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import LayerNorm, SoftmaxXent
+    from shallowspeed_amd.parallel import NaiveParallelSchedule, Topology, Worker
+
+    gbs = 32
+    model = Sequential([
+        Linear(20, 32, activation="gelu"),
+        LayerNorm(32),
+        Linear(32, 10),
+        SoftmaxXent(gbs),
+    ])
+    model.in_dim, model.out_dim = 20, 10
+    model.materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=0.1)
+    ds = Dataset(gbs, 8, n_samples=256, in_dim=20, n_classes=10).load(0, 1)
+    w = Worker(Topology(), model, ds, opt)
+
+    def acc():
+        model.eval()
+        p = model.forward(ds.x, 0)
+        model.train()
+        return (p.argmax(-1) == ds.y.argmax(-1)).float().mean().item()
+
+    a0 = acc()
+    for _ in range(25):
+        for b in range(ds.num_batches()):
+            w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
+    assert acc() > a0 + 0.15

@@ -17,8 +17,6 @@ RCCL/xGMI; the DP all-reduce rides torch.distributed's comm stream and
 overlaps backward kernels.
 """
 
-from typing import Optional
-
 import torch
 
 from ..utils import StepTimer

@@ -248,8 +248,8 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     constexpr int FM = BMT / 32;   // 16x16 frags per wave (2x2 waves)
     constexpr int FN = BNT / 32;
 
-    __shared__ ushort At[BKW / 4 * NMA * 64];  // dY^T tile, blocked
-    __shared__ ushort Bt[BKW / 4 * NMB * 64];  // X^T tile, blocked
+    __shared__ ushort At[BKW / 4 * NMA * 72];  // dY^T tile, blocked+padded
+    __shared__ ushort Bt[BKW / 4 * NMB * 72];  // X^T tile, blocked+padded
     __shared__ float dbs[256 / BMT > 1 ? 256 / BMT : 2][BMT];
 
     const int tid = threadIdx.x;
@@ -271,11 +271,13 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const bool do_db = (gb != nullptr) && (blockIdx.y == 0);
     float db_part = 0.f;
 
+    // subtiles padded 64->72 ushorts (144 B): consecutive k-subtiles no
+    // longer alias the same LDS bank row for the tr-read lane groups
     auto baddrA = [](int k, int m) {
-        return ((k >> 2) * NMA + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
+        return ((k >> 2) * NMA + (m >> 4)) * 72 + (k & 3) * 16 + (m & 15);
     };
     auto baddrB = [](int k, int m) {
-        return ((k >> 2) * NMB + (m >> 4)) * 64 + (k & 3) * 16 + (m & 15);
+        return ((k >> 2) * NMB + (m >> 4)) * 72 + (k & 3) * 16 + (m & 15);
     };
 
     // staging: thread t covers ELA contiguous elems of the [BKW][BMT]
@@ -361,9 +363,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int i = 0; i < FM; ++i) {
                 const int mblk = wm * FM + i;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[(g2 * NMA + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[(g2 * NMA + mblk) * 72 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[((g2 + 1) * NMA + mblk) * 64 + lcol4]);
+                    (lds_v4p)&At[((g2 + 1) * NMA + mblk) * 72 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     a_frag[i][e] = lo[e];
@@ -374,9 +376,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int j = 0; j < FN; ++j) {
                 const int nblk = wn * FN + j;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 72 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 64 + lcol4]);
+                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 72 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     b_frag[j][e] = lo[e];
@@ -510,7 +512,7 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, bm) * cdiv(N, bn);
         split_k = 1;
-        while (tiles * split_k < 1024 && split_k * 128 < Kb) split_k *= 2;
+        while (tiles * split_k < 512 && split_k * 128 < Kb) split_k *= 2;
     }
     int k_per_split = cdiv(cdiv(Kb, split_k), 64) * 64;  // BKW=64
     split_k = cdiv(Kb, k_per_split);

@@ -508,11 +508,17 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     // (measured: at 256-wide layers the 64-config's block count wins)
     const bool big = Mo >= 512 && N >= 512;
     const int bm = big ? 128 : 64, bn = big ? 128 : 64;
-    // pick split so that grid covers the CUs (~2-4 blocks/CU)
+    // pick split for (a) ~8 K-steps per block (k_per_split ≈ 512 —
+    // shorter slices pay too much prologue/atomic epilogue, longer
+    // ones tail-imbalance) and (b) at least ~512 blocks to cover the
+    // 256 CUs (measured sweet spots across the bench shapes)
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, bm) * cdiv(N, bn);
-        split_k = 1;
-        while (tiles * split_k < 512 && split_k * 128 < Kb) split_k *= 2;
+        const int by_depth = cdiv(Kb, 512);
+        const int by_cover = cdiv(512, tiles);
+        split_k = by_depth > by_cover ? by_depth : by_cover;
+        const int max_split = cdiv(Kb, 64);
+        if (split_k > max_split) split_k = max_split;
     }
     int k_per_split = cdiv(cdiv(Kb, split_k), 64) * 64;  // BKW=64
     split_k = cdiv(Kb, k_per_split);

@@ -173,14 +173,6 @@ class Linear(Module):
         mask_src = None
         if self.activation == "relu":
             mask_src = self._unstash("y", mubatch_id)
-            if dout.is_cuda:
-                # On GPU, materialize dz = dy ⊙ 1[y>0] ONCE instead of
-                # fusing the mask into both GEMMs' staging: each GEMM
-                # block re-stages its A tile per n-tile, so a fused
-                # mask is RE-READ n-tile times (13× for a 784-wide
-                # layer) — one cheap elementwise pass beats that.
-                dout = F.relu_bwd(dout, mask_src)
-                mask_src = None
         elif self.activation == "gelu":
             z = self._unstash("z", mubatch_id)
             dout = F.gelu_bwd(dout, z)

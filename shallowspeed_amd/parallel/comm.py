@@ -120,18 +120,53 @@ def init_topology(dp: int, pp: int, backend: Optional[str] = None,
 
 
 # ---------------------------------------------------------------- p2p
+#
+# On the production path (one rank per GPU, backend "nccl" == RCCL on
+# ROCm) these are direct device-to-device xGMI transfers.  The gloo
+# backend cannot move CUDA tensors p2p, so a CPU staging fallback keeps
+# the SAME Worker/schedule code testable with several ranks sharing one
+# GPU (or none) — used by the multi-process GPU integration tests.
+
+
+def _needs_cpu_staging(t: torch.Tensor) -> bool:
+    return t.is_cuda and dist.get_backend() == "gloo"
+
+
+class _StagedWork:
+    """Work-handle wrapper keeping the staged CPU copy alive until
+    wait() (isend fallback)."""
+
+    def __init__(self, work, staged):
+        self._work = work
+        self._staged = staged
+
+    def wait(self):
+        self._work.wait()
+        self._staged = None
+
 
 def send_tensor(t: torch.Tensor, dst_rank: int):
+    if _needs_cpu_staging(t):
+        dist.send(t.detach().cpu(), dst=dst_rank)
+        return
     dist.send(t.contiguous(), dst=dst_rank)
 
 
 def isend_tensor(t: torch.Tensor, dst_rank: int):
     """Nonblocking send; caller waits the returned work handle before
     reusing the buffer (double-buffered PP edges)."""
+    if _needs_cpu_staging(t):
+        staged = t.detach().cpu()
+        return _StagedWork(dist.isend(staged, dst=dst_rank), staged)
     return dist.isend(t.contiguous(), dst=dst_rank)
 
 
 def recv_tensor(t: torch.Tensor, src_rank: int):
+    if _needs_cpu_staging(t):
+        tmp = torch.empty(t.shape, dtype=t.dtype, device="cpu")
+        dist.recv(tmp, src=src_rank)
+        t.copy_(tmp, non_blocking=True)
+        return
     dist.recv(t, src=src_rank)
 
 

@@ -55,6 +55,7 @@ class Worker:
         # async p2p bookkeeping: ("in"|"out", idx) -> pending isend work
         self._pending_send = {}
         self.instruction_times = {}  # class name -> seconds (observability)
+        self.comm_stats = {}         # p2p overlap counters (debug/timing)
         self._timing = False
 
         self._DISPATCH = {
@@ -188,6 +189,17 @@ class Worker:
     def _wait_buffer(self, kind, idx):
         h = self._pending_send.pop((kind, idx), None)
         if h is not None:
+            # overlap observability (the comm/compute race guard — the
+            # analog of the reference's by-construction Iallreduce
+            # safety note at pipe.py:313): count how often the wait
+            # actually blocked vs the send having already drained
+            if self._timing:
+                try:
+                    done = h.is_completed()
+                except Exception:
+                    done = False
+                k = "p2p_wait_blocked" if not done else "p2p_wait_free"
+                self.comm_stats[k] = self.comm_stats.get(k, 0) + 1
             h.wait()
 
     def _isend(self, kind, idx, tensor, dst):

@@ -384,3 +384,31 @@ def test_gpu_ln_gelu_stack_trains(gpu_device):
             w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
     torch.cuda.synchronize()
     assert acc() > a0 + 0.15
+
+
+def test_gpu_reference_default_config_trains(gpu_device):
+    """End-to-end with the reference's exact (odd-sized) layer config
+    [784,128,127,126,125,124,123,10] (reference train.py:98) — edge
+    cases: non-multiple-of-64 dims through every kernel."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import GPipeSchedule, Topology, Worker
+
+    sizes = [784, 128, 127, 126, 125, 124, 123, 10]
+    model = MLP(sizes, 0, 1, 128, loss="mse")  # reference loss head
+    model.materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.05)
+    ds = Dataset(128, 32, n_samples=512, in_dim=784, n_classes=10,
+                 device=gpu_device).load(0, 1)
+    w = Worker(Topology(device=gpu_device), model, ds, opt)
+    h0 = [p.data.float().abs().sum().item() for p in model.parameters()]
+    for _ in range(3):
+        for b in range(ds.num_batches()):
+            w.execute(GPipeSchedule(ds.num_mubatches(), 1, 0), b)
+    torch.cuda.synchronize()
+    for p in model.parameters():
+        v = p.data.float()
+        assert torch.isfinite(v).all()
+    # weights moved
+    h1 = [p.data.float().abs().sum().item() for p in model.parameters()]
+    assert any(abs(a - b) > 1e-6 for a, b in zip(h0, h1))

@@ -31,6 +31,9 @@ def main():
                     choices=["naive", "gpipe", "pipedream"])
     ap.add_argument("--layer-sizes", default="784,256,256,256,10")
     ap.add_argument("--loss", default="xent", choices=["xent", "mse"])
+    ap.add_argument("--backend", default=None,
+                    help="torch.distributed backend override "
+                         "(default: nccl on GPU, gloo on CPU)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph step capture (single-GPU only)")
@@ -49,7 +52,7 @@ def main():
     assert dp * pp == n, f"dp({dp})*pp({pp}) != {n}"
 
     device = torch.device(args.device) if args.device else None
-    topo = init_topology(dp, pp, device=device)
+    topo = init_topology(dp, pp, backend=args.backend, device=device)
     device = topo.device
     is_gpu = device.type == "cuda"
 

@@ -77,6 +77,9 @@ def main():
                     help="load x_*.parquet / y_*.npy like the reference "
                          "if present; otherwise synthetic")
     ap.add_argument("--samples", type=int, default=8192)
+    ap.add_argument("--backend", default=None,
+                    help="torch.distributed backend override "
+                         "(default: nccl on GPU, gloo on CPU)")
     ap.add_argument("--device", default=None, help="cpu | cuda")
     ap.add_argument("--save", default=None, help="checkpoint dir to write")
     ap.add_argument("--timing", action="store_true",
@@ -89,7 +92,8 @@ def main():
     assert len(args.layer_sizes) % args.pp == 0, \
         "len(--layer-sizes) must divide into --pp stages (layers.py:242)"
     device = torch.device(args.device) if args.device else None
-    topo = init_topology(args.dp, args.pp, device=device)
+    topo = init_topology(args.dp, args.pp, backend=args.backend,
+                         device=device)
     device = topo.device
 
     # model: stage slice of the full MLP (reference train.py:99-107)

@@ -58,7 +58,11 @@ def main():
 
     sizes = [int(s) for s in args.layer_sizes.split(",")]
     assert args.local_batch % args.mubatches == 0
-    assert len(sizes) % pp == 0, "layer count must divide into pp stages"
+    # PP needs len(sizes) % pp == 0 (stage slicing, layers.py:242):
+    # pad the default config with extra hidden layers if needed
+    while len(sizes) % pp != 0:
+        sizes.insert(1, sizes[1])
+    assert len(sizes) % pp == 0
     global_batch = args.local_batch * dp
     mubatch = args.local_batch // args.mubatches
 

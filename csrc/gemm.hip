@@ -492,10 +492,13 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
     // blocks each (1 block/CU = 4 waves starves latency hiding)
     if (N >= 128 && (long)cdiv(M, 128) * cdiv(N, 128) >= 512) {
         DISPATCH(B128)
-    } else if (M >= 48) {
-        DISPATCH(B64)
-    } else {
+    } else if (M < 48 ||
+               (long)cdiv(M, 64) * cdiv(N, 64) < 192) {
+        // small-M / small-grid shapes: the 32-row config doubles the
+        // block count (latency-bound regime — µbatched schedules)
         DISPATCH(B32)
+    } else {
+        DISPATCH(B64)
     }
 #undef DISPATCH
 }

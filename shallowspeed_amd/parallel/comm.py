@@ -186,9 +186,16 @@ class GradReducer:
     post-grad hook, before OptimizerStep.
     """
 
-    def __init__(self, model, group, bucket_bytes: int = 25 << 20):
+    def __init__(self, model, group, bucket_bytes=None):
         self.group = group
         self.flat = model._flat_grad
+        if bucket_bytes is None:
+            # adaptive: ~4 buckets per stage so earlier buckets'
+            # all-reduces overlap the remaining backward compute, but
+            # never below 256 KB (latency-bound regime on xGMI) nor
+            # above 25 MB (DDP-style cap, reference pipe.py:309-310)
+            total = self.flat.numel() * 4
+            bucket_bytes = max(256 << 10, min(25 << 20, total // 4))
         self.enabled = group is not None and dist.is_initialized() \
             and dist.get_world_size(group=group) > 1
 

@@ -39,7 +39,7 @@ from .instructions import (
 
 class Worker:
     def __init__(self, topo: Topology, model, dataset=None, optimizer=None,
-                 use_dp: bool = True, bucket_bytes: int = 25 << 20):
+                 use_dp: bool = True, bucket_bytes=None):
         self.topo = topo
         self.model = model
         self.dataset = dataset

@@ -1,0 +1,52 @@
+"""Guards the driver contract: bench.py must print ONE JSON line with
+the required fields, and utils must behave (rank-0 print, hashing)."""
+
+import json
+import subprocess
+import sys
+
+import torch
+
+from shallowspeed_amd.models import MLP
+from shallowspeed_amd.utils import get_model_hash
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "3", "--warmup", "1",
+         "--local-batch", "256", "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, check=True,
+    ).stdout.strip().splitlines()
+    # exactly one JSON line on stdout
+    payload = [l for l in out if l.startswith("{")]
+    assert len(payload) == 1, out
+    j = json.loads(payload[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in j, key
+    assert j["metric"].startswith("samples/sec")
+    assert j["unit"] == "samples/sec"
+    assert j["n_gpus"] == 1 and j["steps"] == 3 and j["warmup"] == 1
+    assert j["higher_is_better"] is True
+    assert j["scaling"] == "weak"
+    assert j["data"] == "synthetic"
+    assert j["value"] > 0 and j["ms_per_step"] > 0
+    for key in ("model", "global_batch", "parallelism"):
+        assert key in j["config"], key
+    assert j["config"]["global_batch"] == 256
+
+
+def test_model_hash_detects_changes():
+    m1 = MLP([16, 8, 4], 0, 1, 8)
+    m2 = MLP([16, 8, 4], 0, 1, 8)
+    assert get_model_hash(m1) == get_model_hash(m2)  # shape-seeded init
+    m2.parameters()[0].data += 1e-3
+    assert get_model_hash(m1) != get_model_hash(m2)
+
+
+def test_rprint_single_process(capsys):
+    from shallowspeed_amd.utils import rprint
+
+    rprint("hello")
+    assert "hello" in capsys.readouterr().out

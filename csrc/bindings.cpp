@@ -23,7 +23,8 @@ void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
                      hipStream_t);
 void ss_head_xent_bwd(const void*, const void*, void*, long, float,
                       hipStream_t);
-void ss_sgd_multi(const void*, int, long, float, hipStream_t);
+void ss_sgd_multi(const void*, int, long, float, float, float,
+                  hipStream_t);
 void ss_ln_fwd(const void*, const void*, const void*, void*, void*, void*,
                int, int, float, hipStream_t);
 void ss_ln_bwd_dx(const void*, const void*, const void*, const void*,
@@ -238,14 +239,15 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor z) {
     return dz;
 }
 
-void sgd_multi(torch::Tensor desc, double lr, int64_t total) {
+void sgd_multi(torch::Tensor desc, double lr, int64_t total,
+               double momentum, double weight_decay) {
     // total passed by the caller (cached host-side) — no device sync.
     TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
                     desc.is_contiguous() && desc.dim() == 2 &&
-                    desc.size(1) == 7,
-                "desc must be CUDA int64 [T,7]");
+                    desc.size(1) == 8,
+                "desc must be CUDA int64 [T,8]");
     ss_sgd_multi(desc.data_ptr(), (int)desc.size(0), (long)total, (float)lr,
-                 cur_stream());
+                 (float)momentum, (float)weight_decay, cur_stream());
 }
 
 }  // namespace

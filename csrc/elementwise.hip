@@ -162,9 +162,13 @@ __global__ __launch_bounds__(256) void head_xent_bwd_kernel(
 
 // --------------------------------------------------------- fused SGD
 
-// desc row (int64 x 7): master*, grad*, lp*, lpt*, numel, cols, start
+// desc row (int64 x 8): master*, grad*, lp*, lpt*, numel, cols, start,
+// velocity* (0 when momentum is off).  One launch updates every stage
+// parameter: master -= lr * (grad + wd*master [+ momentum]), and
+// re-emits the bf16 compute copy and the transposed bf16 copy.
 __global__ __launch_bounds__(256) void sgd_multi_kernel(
-    const long* __restrict__ desc, int ntensors, long total, float lr) {
+    const long* __restrict__ desc, int ntensors, long total, float lr,
+    float momentum, float weight_decay) {
     long e = (long)blockIdx.x * 256 + threadIdx.x;
     const long stride = (long)gridDim.x * 256;
     for (; e < total; e += stride) {
@@ -172,14 +176,21 @@ __global__ __launch_bounds__(256) void sgd_multi_kernel(
         int lo = 0, hi = ntensors - 1;
         while (lo < hi) {
             const int mid = (lo + hi + 1) >> 1;
-            if (desc[mid * 7 + 6] <= e) lo = mid;
+            if (desc[mid * 8 + 6] <= e) lo = mid;
             else hi = mid - 1;
         }
-        const long* d = desc + lo * 7;
+        const long* d = desc + lo * 8;
         const long i = e - d[6];
         float* master = (float*)d[0];
         const float* grad = (const float*)d[1];
-        const float m = master[i] - lr * grad[i];
+        float g = grad[i];
+        if (weight_decay != 0.f) g += weight_decay * master[i];
+        if (d[7]) {
+            float* vel = (float*)d[7];
+            g = momentum * vel[i] + g;
+            vel[i] = g;
+        }
+        const float m = master[i] - lr * g;
         master[i] = m;
         const __bf16 v = f2bf(m);
         if (d[2]) ((__bf16*)d[2])[i] = v;
@@ -239,11 +250,12 @@ void ss_head_xent_bwd(const void* s, const void* t, void* dz, long n,
 }
 
 void ss_sgd_multi(const void* desc, int ntensors, long total, float lr,
-                  hipStream_t st) {
+                  float momentum, float weight_decay, hipStream_t st) {
     long blocks = (total + 255) / 256;
     if (blocks > 1024) blocks = 1024;
     hipLaunchKernelGGL(sgd_multi_kernel, dim3((int)blocks), dim3(256), 0, st,
-                       (const long*)desc, ntensors, total, lr);
+                       (const long*)desc, ntensors, total, lr, momentum,
+                       weight_decay);
 }
 
 }  // extern "C"

@@ -412,3 +412,24 @@ def test_gpu_reference_default_config_trains(gpu_device):
     # weights moved
     h1 = [p.data.float().abs().sum().item() for p in model.parameters()]
     assert any(abs(a - b) > 1e-6 for a, b in zip(h0, h1))
+
+
+def test_gpu_sgd_momentum_weight_decay(gpu_device):
+    from shallowspeed_amd.models import Linear, SGD, Sequential
+
+    model = Sequential([Linear(65, 33), Linear(33, 10)])
+    model.materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=0.01)
+    tw = [p.data.clone().requires_grad_(True) for p in model.parameters()]
+    topt = torch.optim.SGD(tw, lr=0.1, momentum=0.9, weight_decay=0.01)
+    for step in range(3):
+        for p, t in zip(model.parameters(), tw):
+            g = torch.randn_like(p.data)
+            p.grad.copy_(g)
+            t.grad = g.clone()
+        opt.step()
+        topt.step()
+    torch.cuda.synchronize()
+    for p, t in zip(model.parameters(), tw):
+        torch.testing.assert_close(p.data, t.detach(), atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(p.lp.float(), p.data.to(BF).float())

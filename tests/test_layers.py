@@ -254,3 +254,20 @@ def test_layernorm_mlp_trains():
         for b in range(ds.num_batches()):
             w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
     assert acc() > a0 + 0.15
+
+
+def test_sgd_momentum_weight_decay_matches_torch():
+    """Fused-SGD semantics (momentum + weight decay) vs torch.optim.SGD."""
+    model = Sequential([Linear(6, 5), Linear(5, 4)])
+    opt = SGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=0.01)
+    tw = [p.data.clone().requires_grad_(True) for p in model.parameters()]
+    topt = torch.optim.SGD(tw, lr=0.1, momentum=0.9, weight_decay=0.01)
+    for step in range(3):
+        for p, t in zip(model.parameters(), tw):
+            g = torch.randn_like(p.data)
+            p.grad.copy_(g)
+            t.grad = g.clone()
+        opt.step()
+        topt.step()
+    for p, t in zip(model.parameters(), tw):
+        torch.testing.assert_close(p.data, t.detach(), atol=1e-6, rtol=1e-5)

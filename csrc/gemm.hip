@@ -509,7 +509,7 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     // big tiles only pay when both dims are wide enough that the
     // grid still covers the CUs AND operand re-reads dominate
     // (measured: at 256-wide layers the 64-config's block count wins)
-    const bool big = Mo >= 512 && N >= 512;
+    const bool big = (Mo >= 256 && N >= 512) || (Mo >= 512 && N >= 256);
     const int bm = big ? 128 : 64, bn = big ? 128 : 64;
     // pick split for (a) ~8 K-steps per block (k_per_split ≈ 512 —
     // shorter slices pay too much prologue/atomic epilogue, longer

@@ -44,9 +44,9 @@ __device__ __forceinline__ int kswz(int m, int k) {
 // (registers -> padded LDS tile) so the HBM latency of tile t+1 hides
 // under tile t's MFMA work (async-STAGE split, one barrier per K-step;
 // cdna_hip_programming.md Guideline 15 / T14).
-template <int TILE_ROWS, bool HAS_MASK>
+template <int TILE_ROWS, int KSTEP, bool HAS_MASK>
 struct StageReg {
-    static constexpr int EL = TILE_ROWS * BK / 256;  // 4, 8 or 16
+    static constexpr int EL = TILE_ROWS * KSTEP / 256;  // 4..32
     __bf16 v[EL];
 
     __device__ __forceinline__ void load(const __bf16* __restrict__ src,
@@ -54,7 +54,7 @@ struct StageReg {
                                          int nrows, int K, int row0, int k0,
                                          int tid) {
         const int off = tid * EL;
-        const int r = off / BK, c = off % BK;
+        const int r = off / KSTEP, c = off % KSTEP;
         const int g = row0 + r, gk = k0 + c;
         if (g < nrows && gk + EL <= K) {
 #pragma unroll
@@ -100,9 +100,9 @@ struct StageReg {
     }
 
     __device__ __forceinline__ void write(
-        ushort (*__restrict__ dst)[BK + LDS_PAD], int tid) const {
+        ushort (*__restrict__ dst)[KSTEP + LDS_PAD], int tid) const {
         const int off = tid * EL;
-        const int r = off / BK, c = off % BK;
+        const int r = off / KSTEP, c = off % KSTEP;
 #pragma unroll
         for (int ch = 0; ch < EL; ch += (EL < 8 ? 4 : 8)) {
             if constexpr (EL >= 8) {
@@ -120,7 +120,7 @@ struct StageReg {
     }
 };
 
-template <int BM, int BN, int WAVES_M, int WAVES_N,
+template <int BM, int BN, int KSTEP, int WAVES_M, int WAVES_N,
           bool HAS_BIAS, bool RELU, bool HAS_MASK>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const __bf16* __restrict__ A,     // [M][K]
@@ -133,7 +133,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     constexpr int WN = BN / WAVES_N;        // wave tile cols
     constexpr int FM = WM / 16;             // 16x16 frags per wave (M)
     constexpr int FN = WN / 16;
-    constexpr int LDA = BK + LDS_PAD;
+    constexpr int LDA = KSTEP + LDS_PAD;
 
     __shared__ ushort As[2][BM][LDA];
     __shared__ ushort Bs[2][BN][LDA];
@@ -155,40 +155,43 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const int lrow = lane & 15;
     const int kch = lane >> 4;  // 0..3 -> k offset kch*8
 
-    StageReg<BM, HAS_MASK> ra;
-    StageReg<BN, false> rb;
+    StageReg<BM, KSTEP, HAS_MASK> ra;
+    StageReg<BN, KSTEP, false> rb;
     ra.load(A, mask, M, K, m0, 0, tid);
     rb.load(B, nullptr, N, K, n0, 0, tid);
     ra.write(As[0], tid);
     rb.write(Bs[0], tid);
     __syncthreads();
 
-    const int nsteps = (K + BK - 1) / BK;
+    const int nsteps = (K + KSTEP - 1) / KSTEP;
     int cur = 0;
     for (int t = 0; t < nsteps; ++t) {
         // issue next tile's global loads early (latency hides under
         // this tile's MFMA)
         if (t + 1 < nsteps) {
-            ra.load(A, mask, M, K, m0, (t + 1) * BK, tid);
-            rb.load(B, nullptr, N, K, n0, (t + 1) * BK, tid);
+            ra.load(A, mask, M, K, m0, (t + 1) * KSTEP, tid);
+            rb.load(B, nullptr, N, K, n0, (t + 1) * KSTEP, tid);
         }
 
         // ---- MFMA on the current tile ----
-        bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
-        for (int i = 0; i < FM; ++i)
-            a_frag[i] =
-                *(const bf16x8*)&As[cur][wm * WM + i * 16 + lrow][kch * 8];
+        for (int kk = 0; kk < KSTEP / 32; ++kk) {
+            bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
-        for (int j = 0; j < FN; ++j)
-            b_frag[j] =
-                *(const bf16x8*)&Bs[cur][wn * WN + j * 16 + lrow][kch * 8];
-#pragma unroll
-        for (int i = 0; i < FM; ++i)
+            for (int i = 0; i < FM; ++i)
+                a_frag[i] = *(const bf16x8*)&As[cur][wm * WM + i * 16 + lrow]
+                                                   [kk * 32 + kch * 8];
 #pragma unroll
             for (int j = 0; j < FN; ++j)
-                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+                b_frag[j] = *(const bf16x8*)&Bs[cur][wn * WN + j * 16 + lrow]
+                                                   [kk * 32 + kch * 8];
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        }
 
         if (t + 1 < nsteps) {
             ra.write(As[cur ^ 1], tid);
@@ -468,11 +471,23 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
         constexpr int WAVES_M = BM == 32 ? 1 : 2;
         constexpr int WAVES_N = BM == 32 ? 4 : 2;
         dim3 grid(cdiv(M, BM), cdiv(N, BN));
-        hipLaunchKernelGGL(
-            (gemm_nt_kernel<BM, BN, WAVES_M, WAVES_N, HB, RL, HM>),
-            grid, block, 0, stream,
-            (const __bf16*)A, (const __bf16*)B, (const __bf16*)bias,
-            (const __bf16*)mask, (__bf16*)C, M, N, K);
+        // Measured: KSTEP=64 is throughput-neutral at these shapes
+        // (barrier savings offset by the occupancy drop 7->4
+        // blocks/CU), so the deep K-step stays off; the template
+        // keeps it one constant away for wider models.
+        if (false && K >= 512 && BM >= 64) {
+            hipLaunchKernelGGL(
+                (gemm_nt_kernel<BM, BN, 64, WAVES_M, WAVES_N, HB, RL, HM>),
+                grid, block, 0, stream,
+                (const __bf16*)A, (const __bf16*)B, (const __bf16*)bias,
+                (const __bf16*)mask, (__bf16*)C, M, N, K);
+        } else {
+            hipLaunchKernelGGL(
+                (gemm_nt_kernel<BM, BN, 32, WAVES_M, WAVES_N, HB, RL, HM>),
+                grid, block, 0, stream,
+                (const __bf16*)A, (const __bf16*)B, (const __bf16*)bias,
+                (const __bf16*)mask, (__bf16*)C, M, N, K);
+        }
     };
 
     using T = std::true_type;

@@ -59,6 +59,9 @@ class Schedule:
     def is_last_mubatch(self, mubatch_id):
         return mubatch_id == self.num_micro_batches - 1
 
+    # training schedules run backward; InferenceSchedule overrides.
+    is_training = True
+
     # buffer assignment: µbatch m round-robins over the buffer pairs.
     # Sends are ASYNC (the Worker isends and waits only when a buffer
     # is about to be overwritten), so schedules that keep >1 µbatch in
@@ -221,6 +224,8 @@ class PipeDreamFlushSchedule(Schedule):
 class InferenceSchedule(Schedule):
     """Forward-only pipeline for eval.  Reference: pipe.py:275-294
     (used by compute_accuracy, train.py:32-37)."""
+
+    is_training = False
 
     def steps(self):
         for m in range(self.num_micro_batches):

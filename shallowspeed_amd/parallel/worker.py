@@ -103,6 +103,16 @@ class Worker:
     def execute(self, schedule, batch_id: int):
         self._batch_id = batch_id
         self._snapshot_inputs = getattr(schedule, "max_in_flight", 2) > 1
+        # the last training stage's forward output is consumed only via
+        # the head's stash (the out buffer would be dead weight: it is
+        # overwritten by LoadMuBatchTarget into the grad-in pool) —
+        # skip the staging copy there
+        self._skip_out_copy = (
+            getattr(schedule, "is_training", True)
+            and self.model._training
+            and self.topo.stage_id == self.topo.pp - 1
+        )
+        self._in_views = {}
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
@@ -224,9 +234,16 @@ class Worker:
             x = self._staged_x[cmd.mubatch_id * mb:(cmd.mubatch_id + 1) * mb]
         else:
             x = self.dataset.micro_batch_input(self._batch_id, cmd.mubatch_id)
-        self._wait_buffer("in", cmd.buffer_idx)
         buf = self._in_bufs[cmd.buffer_idx]
         assert x.shape == buf.shape, (x.shape, buf.shape)  # pipe.py:357-360
+        if x.dtype == self.compute_dtype and x.device == buf.device:
+            # zero-copy: the forward reads the dataset slice directly
+            # (the staging buffer exists for the RECV path; stage 0
+            # never receives).  The slice is read-only to the model.
+            self._in_views[cmd.buffer_idx] = x
+            return
+        self._wait_buffer("in", cmd.buffer_idx)
+        self._in_views.pop(cmd.buffer_idx, None)
         buf.copy_(x.to(self.compute_dtype), non_blocking=True)
 
     def _load_target(self, cmd):
@@ -241,6 +258,7 @@ class Worker:
         buf.copy_(y.to(self.compute_dtype), non_blocking=True)
 
     def _recv_activations(self, cmd):
+        self._in_views.pop(cmd.buffer_idx, None)
         self._wait_buffer("in", cmd.buffer_idx)
         comm_mod.recv_tensor(self._in_bufs[cmd.buffer_idx], self.topo.prev_rank)
 
@@ -257,7 +275,9 @@ class Worker:
                     self._gout_bufs[cmd.buffer_idx], self.topo.prev_rank)
 
     def _forward(self, cmd):
-        x = self._in_bufs[cmd.in_buffer]
+        x = self._in_views.get(cmd.in_buffer)
+        if x is None:
+            x = self._in_bufs[cmd.in_buffer]
         if self.model._training and self._snapshot_inputs:
             # Layers stash their input per µbatch for wgrad; the input
             # buffer is SHARED across µbatches (overwritten by the next
@@ -267,6 +287,8 @@ class Worker:
             # only safe for its naive schedule ordering.)
             x = x.clone()
         y = self.model.forward(x, cmd.mubatch_id)
+        if getattr(self, "_skip_out_copy", False):
+            return
         self._wait_buffer("out", cmd.out_buffer)
         self._out_bufs[cmd.out_buffer].copy_(y)
 

@@ -260,9 +260,23 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const int wave = tid >> 6;
     const int wm = wave >> 1;  // 2x2 waves
     const int wn = wave & 1;
-    const int m0 = blockIdx.x * BMT;
-    const int n0 = blockIdx.y * BNT;
-    const int kbeg = blockIdx.z * k_per_split;
+    // XCD-aware block remap (T1, bijective variant): the dispatcher
+    // places hw block b on XCD b%8; remap so each XCD owns a
+    // CONTIGUOUS run of work ids decoded n-tile-fastest — blocks that
+    // share a dY m-slice then read it from ONE XCD's L2 instead of
+    // filling all eight.
+    const int gx = gridDim.x, gy = gridDim.y;
+    const int nwg = gx * gy * gridDim.z;
+    const int hw = blockIdx.x + gx * (blockIdx.y + gy * blockIdx.z);
+    const int xcd = hw % 8, q = nwg / 8, r = nwg % 8;
+    const int wid = (xcd < r ? xcd * (q + 1)
+                             : r * (q + 1) + (xcd - r) * q) + hw / 8;
+    const int bidy = wid % gy;
+    const int bidx = (wid / gy) % gx;
+    const int bidz = wid / (gy * gx);
+    const int m0 = bidx * BMT;
+    const int n0 = bidy * BNT;
+    const int kbeg = bidz * k_per_split;
     const int kend = min(Kb, kbeg + k_per_split);
 
     f32x4 acc[FM][FN];
@@ -271,7 +285,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #pragma unroll
         for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const bool do_db = (gb != nullptr) && (blockIdx.y == 0);
+    const bool do_db = (gb != nullptr) && (bidy == 0);
     float db_part = 0.f;
 
     // subtiles padded 64->72 ushorts (144 B): consecutive k-subtiles no

@@ -143,8 +143,17 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const int wave = tid >> 6;
     const int wm = wave / WAVES_N;
     const int wn = wave % WAVES_N;
-    const int m0 = blockIdx.x * BM;
-    const int n0 = blockIdx.y * BN;
+    // XCD-aware bijective remap (T1): decode the work id n-tile-fastest
+    // so the blocks sharing an A (activation) m-slice run on ONE XCD
+    // and hit its L2 instead of filling all eight.
+    const int gx = gridDim.x, gy = gridDim.y;
+    const int nwg = gx * gy;
+    const int hw = blockIdx.x + gx * blockIdx.y;
+    const int xcd = hw % 8, q8 = nwg / 8, r8 = nwg % 8;
+    const int wid = (xcd < r8 ? xcd * (q8 + 1)
+                              : r8 * (q8 + 1) + (xcd - r8) * q8) + hw / 8;
+    const int m0 = ((wid / gy) % gx) * BM;
+    const int n0 = (wid % gy) * BN;
 
     f32x4 acc[FM][FN];
 #pragma unroll

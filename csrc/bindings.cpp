@@ -14,6 +14,8 @@ void ss_gemm_nt(const void*, const void*, const void*, const void*, void*,
                 int, int, int, bool, hipStream_t);
 void ss_wgrad_tn(const void*, const void*, const void*, void*, void*, int,
                  int, int, int, hipStream_t);
+void ss_wgrad_tn_multi(const void*, int, bool, void*, void*, int, int, int,
+                       int, hipStream_t);
 void ss_colsum(const void*, const void*, void*, int, int, hipStream_t);
 void ss_relu_fwd(const void*, void*, long, hipStream_t);
 void ss_relu_bwd(const void*, const void*, void*, long, hipStream_t);
@@ -105,6 +107,32 @@ void wgrad_tn(torch::Tensor dy, torch::Tensor x, torch::Tensor gw,
     // extra dY read on the hot path.
     ss_wgrad_tn(dy.data_ptr(), x.data_ptr(), mask_p, gw.data_ptr(), gb_p, Mo,
                 N, Kb, (int)split_k, cur_stream());
+}
+
+void wgrad_tn_multi(torch::Tensor chunk_table, int64_t nchunks,
+                    bool has_mask, torch::Tensor gw, torch::Tensor gb,
+                    int64_t Mo, int64_t N, int64_t kb_chunk,
+                    int64_t split_k) {
+    // chunk_table: CUDA int64 [nchunks, 3] = {dY*, X*, mask*} per
+    // µbatch; all chunks share (Kb, Mo, N).  Deferred-µbatch wgrad:
+    // one launch accumulates every chunk into gw/gb.
+    TORCH_CHECK(chunk_table.is_cuda() &&
+                    chunk_table.scalar_type() == torch::kLong &&
+                    chunk_table.is_contiguous() &&
+                    chunk_table.dim() == 2 && chunk_table.size(1) == 3 &&
+                    chunk_table.size(0) == nchunks,
+                "chunk_table must be CUDA int64 [nchunks,3]");
+    check_f32(gw, "gw");
+    TORCH_CHECK(gw.size(0) == Mo && gw.size(1) == N, "gw shape");
+    void* gb_p = nullptr;
+    if (has(gb)) {
+        check_f32(gb, "gb");
+        TORCH_CHECK(gb.numel() == Mo, "gb size");
+        gb_p = gb.data_ptr();
+    }
+    ss_wgrad_tn_multi(chunk_table.data_ptr(), (int)nchunks, has_mask,
+                      gw.data_ptr(), gb_p, (int)Mo, (int)N, (int)kb_chunk,
+                      (int)split_k, cur_stream());
 }
 
 torch::Tensor colsum(torch::Tensor dy, torch::Tensor mask) {
@@ -256,6 +284,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
     m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
     m.def("colsum", &colsum, "standalone column sum (bias grad)");
+    m.def("wgrad_tn_multi", &wgrad_tn_multi,
+          "chunked (deferred-µbatch) wgrad: one launch, many (dy,x) pairs");
     m.def("relu_fwd", &relu_fwd);
     m.def("relu_bwd", &relu_bwd);
     m.def("softmax_fwd", &softmax_fwd);

@@ -241,7 +241,12 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const __bf16* __restrict__ mask,  // [Kb][Mo]
     float* __restrict__ gW,           // [Mo][N] (atomicAdd +=)
     float* __restrict__ gb,           // [Mo] or null: fused bias grad
-    int Mo, int N, int Kb, int k_per_split) {
+    int Mo, int N, int Kb, int k_per_split,
+    // optional CHUNK TABLE (deferred µbatch wgrad): int64 rows of
+    // {dY*, X*, mask*}; grid.z covers chunks*split and each chunk is
+    // an independent [Kb][*] pair accumulated into the same gW/gb —
+    // ONE launch replaces num_µbatches launches per layer.
+    const long* __restrict__ chunks, int nsplit) {
     // TN GEMM over the batch axis.  Tiles are staged in a BLOCKED
     // [BKW/4][rows/16][4][16] bf16 layout so that
     //   * staging is plain 16-B vector writes (global rows are
@@ -282,7 +287,14 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
                              : r * (q + 1) + (xcd - r) * q) + hw / 8;
     const int bidy = wid % gy;
     const int bidx = (wid / gy) % gx;
-    const int bidz = wid / (gy * gx);
+    int bidz = wid / (gy * gx);
+    if (chunks) {
+        const long* row = chunks + (long)(bidz / nsplit) * 3;
+        dY = (const __bf16*)row[0];
+        X = (const __bf16*)row[1];
+        mask = (const __bf16*)row[2];
+        bidz = bidz % nsplit;
+    }
     const int m0 = bidx * BMT;
     const int n0 = bidy * BNT;
     const int kbeg = bidz * k_per_split;
@@ -569,13 +581,44 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     hipLaunchKernelGGL((wgrad_tn_kernel<BMT, BNT, HM>), grid, block, 0,     \
                        stream, (const __bf16*)dY, (const __bf16*)X,         \
                        (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,  \
-                       Kb, k_per_split)
+                       Kb, k_per_split, (const long*)nullptr, 1)
     if (big) {
         if (mask) WLAUNCH(128, 128, true); else WLAUNCH(128, 128, false);
     } else {
         if (mask) WLAUNCH(64, 64, true); else WLAUNCH(64, 64, false);
     }
 #undef WLAUNCH
+}
+
+void ss_wgrad_tn_multi(const void* chunk_table, int nchunks, bool has_mask,
+                       void* gW, void* gb, int Mo, int N, int Kb_chunk,
+                       int split_k, hipStream_t stream) {
+    const bool big = (Mo >= 256 && N >= 512) || (Mo >= 512 && N >= 256);
+    const int bm = big ? 128 : 64, bn = big ? 128 : 64;
+    if (split_k <= 0) {
+        const int tiles = cdiv(Mo, bm) * cdiv(N, bn) * nchunks;
+        const int by_depth = cdiv(Kb_chunk, 512);
+        const int by_cover = cdiv(512, tiles);
+        split_k = by_depth > by_cover ? by_depth : by_cover;
+        const int max_split = cdiv(Kb_chunk, 64);
+        if (split_k > max_split) split_k = max_split;
+    }
+    int k_per_split = cdiv(cdiv(Kb_chunk, split_k), 64) * 64;
+    split_k = cdiv(Kb_chunk, k_per_split);
+    dim3 grid(cdiv(Mo, bm), cdiv(N, bn), split_k * nchunks);
+    dim3 block(256);
+#define WLAUNCHM(BMT, BNT, HM)                                              \
+    hipLaunchKernelGGL((wgrad_tn_kernel<BMT, BNT, HM>), grid, block, 0,     \
+                       stream, (const __bf16*)nullptr,                      \
+                       (const __bf16*)nullptr, (const __bf16*)nullptr,      \
+                       (float*)gW, (float*)gb, Mo, N, Kb_chunk,             \
+                       k_per_split, (const long*)chunk_table, split_k)
+    if (big) {
+        if (has_mask) WLAUNCHM(128, 128, true); else WLAUNCHM(128, 128, false);
+    } else {
+        if (has_mask) WLAUNCHM(64, 64, true); else WLAUNCHM(64, 64, false);
+    }
+#undef WLAUNCHM
 }
 
 void ss_colsum(const void* dY, const void* mask, void* gb, int M, int N,

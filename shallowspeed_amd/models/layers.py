@@ -149,6 +149,10 @@ class Linear(Module):
         self._params["bias"] = Parameter(torch.zeros(out_dims))
         self.weight = self._params["weight"]
         self.bias = self._params["bias"]
+        # deferred-µbatch wgrad (GPU): collect (dy, x, mask) per
+        # µbatch and launch ONE chunked kernel at flush time
+        self._defer_wgrad = False
+        self._wgrad_pending = []
 
     def forward(self, inputs, mubatch_id: int = 0):
         self._stash("x", mubatch_id, inputs)
@@ -180,8 +184,20 @@ class Linear(Module):
         if need_dx:
             dx = F.linear_dgrad(dout, self.weight.compute(),
                                 self.weight.compute_t(), mask_src)
-        F.linear_wgrad_acc(dout, x, self.weight.grad, self.bias.grad, mask_src)
+        if self._defer_wgrad and dout.is_cuda:
+            self._wgrad_pending.append((dout, x, mask_src))
+        else:
+            F.linear_wgrad_acc(dout, x, self.weight.grad, self.bias.grad,
+                               mask_src)
         return dx
+
+    def flush_wgrad(self):
+        """Launch the deferred µbatch weight gradients as ONE chunked
+        kernel (see functional.linear_wgrad_multi)."""
+        if self._wgrad_pending:
+            F.linear_wgrad_multi(self._wgrad_pending, self.weight.grad,
+                                 self.bias.grad)
+            self._wgrad_pending = []
 
 
 class ReLU(Module):
@@ -398,6 +414,25 @@ class Sequential(Module):
         self._training = False
         for l in self.layers:
             l.eval()
+
+    def set_defer_wgrad(self, flag: bool):
+        """Enable/disable deferred-µbatch wgrad on every Linear
+        (pipeline schedules with >1 µbatch; flushed before the
+        optimizer step)."""
+        for layer in self.layers:
+            if hasattr(layer, "_defer_wgrad"):
+                layer._defer_wgrad = flag
+
+    def flush_wgrads(self, per_layer_hook=None):
+        """Flush deferred weight gradients layer by layer in BACKWARD
+        order (so DP bucket all-reduces of later layers overlap the
+        remaining wgrad kernels); per_layer_hook(layer) fires after
+        each layer's grads are final."""
+        for layer in reversed(self.layers):
+            if hasattr(layer, "flush_wgrad"):
+                layer.flush_wgrad()
+            if per_layer_hook is not None:
+                per_layer_hook(layer)
 
     def materialize_device(self, device, compute_dtype=None):
         """Move to device, set compute dtype, and re-point every

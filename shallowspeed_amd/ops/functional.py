@@ -232,3 +232,34 @@ def gelu_bwd(dy, z):
     y = torch.nn.functional.gelu(zf, approximate="tanh")
     y.backward(dy.float())
     return zf.grad.to(dy.dtype)
+
+
+def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
+    """Deferred-µbatch weight gradients: ONE kernel launch accumulates
+    every (dy, x[, mask]) chunk of a layer into grad_w/grad_b.
+
+    chunks: list of (dy, x, mask_or_None) with identical shapes.
+    Replaces num_µbatches separate wgrad launches under pipeline
+    schedules (each launch re-pays prologue + atomic epilogue).
+    GPU only; the CPU path loops (it has no launch cost to amortize).
+    """
+    dy0 = chunks[0][0]
+    if not _is_gpu(dy0):
+        for dy, x, m in chunks:
+            linear_wgrad_acc(dy, x, grad_w, grad_b, m, split_k)
+        return
+    ext = _ext_for(dy0)
+    has_mask = chunks[0][2] is not None
+    rows = []
+    for dy, x, m in chunks:
+        assert dy.shape == dy0.shape and (m is not None) == has_mask
+        rows.append([dy.data_ptr(), x.data_ptr(),
+                     m.data_ptr() if m is not None else 0])
+    table = torch.tensor(rows, dtype=torch.int64).to(dy0.device,
+                                                     non_blocking=True)
+    ext.wgrad_tn_multi(table, len(chunks), has_mask, grad_w,
+                       grad_b if grad_b is not None else torch.Tensor(),
+                       dy0.shape[1], chunks[0][1].shape[1], dy0.shape[0],
+                       int(split_k))
+    # keep referenced tensors alive until the kernel has consumed them
+    # (caller holds them in its pending list until flush returns)

@@ -113,6 +113,16 @@ class Worker:
             and self.topo.stage_id == self.topo.pp - 1
         )
         self._in_views = {}
+        # deferred-µbatch wgrad: one chunked kernel per layer at the
+        # optimizer step instead of a wgrad launch per µbatch
+        self._defer_active = (
+            self.device.type == "cuda"
+            and getattr(schedule, "is_training", True)
+            and schedule.num_micro_batches > 1
+            and hasattr(self.model, "set_defer_wgrad")
+        )
+        if hasattr(self.model, "set_defer_wgrad"):
+            self.model.set_defer_wgrad(self._defer_active)
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
@@ -226,7 +236,26 @@ class Worker:
         self.model.zero_grad()
 
     def _optimizer_step(self, cmd):
+        if getattr(self, "_defer_active", False):
+            self._flush_deferred_wgrads()
         self.optimizer.step()
+
+    def _flush_deferred_wgrads(self):
+        """Chunked wgrad flush in backward layer order; in DP mode the
+        bucket all-reduces fire here (grads only become final now) and
+        overlap the remaining layers' wgrad kernels."""
+        if self.reducer is not None:
+            self.reducer.reset()
+
+            def hook(layer):
+                for p in layer.parameters():
+                    if p.requires_grad:
+                        self.reducer.param_done(p)
+
+            self.model.flush_wgrads(hook)
+            self.reducer.finalize()
+        else:
+            self.model.flush_wgrads()
 
     def _load_input(self, cmd):
         if getattr(self, "_use_staged", False):
@@ -303,7 +332,12 @@ class Worker:
     def _backward_and_reduce(self, cmd):
         """Install DP hooks, run backward, reset hooks — the one
         'inversion' in the stack (pipe.py:389-400 ↔ layers.py:201-213):
-        grad-ready → bucket all-reduce launches mid-backward."""
+        grad-ready → bucket all-reduce launches mid-backward.
+
+        In deferred-wgrad mode grads are not final until the flush at
+        OptimizerStep, so the hooks move there."""
+        if getattr(self, "_defer_active", False):
+            return self._backward_acc(cmd)
         if self.reducer is not None:
             self.reducer.reset()
             self.model.register_grad_hook(self.reducer.param_done)

@@ -80,12 +80,7 @@ def main():
     sched = sched_cls(ds.num_mubatches(), pp, topo.stage_id)
     nb = ds.num_batches()
 
-    # hipGraph capture runs the step with per-µbatch eager wgrads
-    # (deferred chunked wgrad cannot build its table under capture);
-    # measured: deferred-eager beats graphed-eager for µbatched
-    # schedules, so graphs engage only for the µ=1 path.
-    use_graph = (not args.no_graph) and is_gpu and topo.world == 1 \
-        and args.mubatches == 1
+    use_graph = (not args.no_graph) and is_gpu and topo.world == 1
 
     def one_step(i):
         if use_graph:

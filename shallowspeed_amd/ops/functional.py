@@ -234,6 +234,9 @@ def gelu_bwd(dy, z):
     return zf.grad.to(dy.dtype)
 
 
+_WGRAD_TABLES = {}
+
+
 def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
     """Deferred-µbatch weight gradients: ONE kernel launch accumulates
     every (dy, x[, mask]) chunk of a layer into grad_w/grad_b.
@@ -250,16 +253,27 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
         return
     ext = _ext_for(dy0)
     has_mask = chunks[0][2] is not None
-    rows = []
-    for dy, x, m in chunks:
+    # Pinned-host staging + cached device table: the H2D becomes an
+    # async (capture-legal) copy, so the flush works INSIDE hipGraph
+    # capture — the pinned buffer stays alive with stable content, and
+    # replays re-copy it (chunk pointers are stable under the graph
+    # memory pool).
+    key = (grad_w.data_ptr(), len(chunks))
+    entry = _WGRAD_TABLES.get(key)
+    if entry is None:
+        pinned = torch.empty(len(chunks), 3, dtype=torch.int64,
+                             pin_memory=True)
+        dev = torch.empty(len(chunks), 3, dtype=torch.int64,
+                          device=dy0.device)
+        entry = _WGRAD_TABLES[key] = (pinned, dev)
+    pinned, table = entry
+    for i, (dy, x, m) in enumerate(chunks):
         assert dy.shape == dy0.shape and (m is not None) == has_mask
-        rows.append([dy.data_ptr(), x.data_ptr(),
-                     m.data_ptr() if m is not None else 0])
-    table = torch.tensor(rows, dtype=torch.int64).to(dy0.device,
-                                                     non_blocking=True)
+        pinned[i, 0] = dy.data_ptr()
+        pinned[i, 1] = x.data_ptr()
+        pinned[i, 2] = m.data_ptr() if m is not None else 0
+    table.copy_(pinned, non_blocking=True)
     ext.wgrad_tn_multi(table, len(chunks), has_mask, grad_w,
                        grad_b if grad_b is not None else torch.Tensor(),
                        dy0.shape[1], chunks[0][1].shape[1], dy0.shape[0],
                        int(split_k))
-    # keep referenced tensors alive until the kernel has consumed them
-    # (caller holds them in its pending list until flush returns)

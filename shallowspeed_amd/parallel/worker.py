@@ -179,11 +179,6 @@ class Worker:
         g = self._graphs.get(key)
         if g is None:
             self._use_staged = not per_batch
-            # deferred wgrad builds its chunk table with a pageable H2D
-            # copy, which ABORTS inside hipGraph capture; a captured
-            # graph amortizes launches anyway, so run graphs eagerly
-            # per µbatch
-            self._force_no_defer = True
             try:
                 # eager warmup (allocates buffers, builds SGD desc) on a
                 # side stream, then capture
@@ -201,7 +196,6 @@ class Worker:
                 return self.execute(schedule, batch_id)
             finally:
                 self._use_staged = False
-                self._force_no_defer = False
             self._graphs[key] = g
         elif g is False:  # capture failed before: stay eager
             return self.execute(schedule, batch_id)

@@ -150,8 +150,14 @@ class Linear(Module):
         self.weight = self._params["weight"]
         self.bias = self._params["bias"]
         # deferred-µbatch wgrad (GPU): collect (dy, x, mask) per
-        # µbatch and launch ONE chunked kernel at flush time
+        # µbatch and launch ONE chunked kernel at flush time.  The
+        # WINDOW caps how many µbatches are retained (holding every
+        # µbatch's dout/x would break 1F1B's bounded activation stash
+        # — the schedule's whole memory story): when the window fills,
+        # the pending chunks flush early (grads accumulate atomically,
+        # so partial flushes compose).
         self._defer_wgrad = False
+        self._wgrad_window = 1 << 30
         self._wgrad_pending = []
 
     def forward(self, inputs, mubatch_id: int = 0):
@@ -186,6 +192,8 @@ class Linear(Module):
                                 self.weight.compute_t(), mask_src)
         if self._defer_wgrad and dout.is_cuda:
             self._wgrad_pending.append((dout, x, mask_src))
+            if len(self._wgrad_pending) >= self._wgrad_window:
+                self.flush_wgrad()
         else:
             F.linear_wgrad_acc(dout, x, self.weight.grad, self.bias.grad,
                                mask_src)
@@ -415,13 +423,16 @@ class Sequential(Module):
         for l in self.layers:
             l.eval()
 
-    def set_defer_wgrad(self, flag: bool):
+    def set_defer_wgrad(self, flag: bool, window: int = 1 << 30):
         """Enable/disable deferred-µbatch wgrad on every Linear
         (pipeline schedules with >1 µbatch; flushed before the
-        optimizer step)."""
+        optimizer step).  window caps retained µbatches — pass the
+        schedule's max_in_flight so 1F1B's activation-memory bound
+        survives deferral."""
         for layer in self.layers:
             if hasattr(layer, "_defer_wgrad"):
                 layer._defer_wgrad = flag
+                layer._wgrad_window = max(1, window)
 
     def flush_wgrads(self, per_layer_hook=None):
         """Flush deferred weight gradients layer by layer in BACKWARD

@@ -113,3 +113,42 @@ def test_pp2_on_gpu_matches_serial(tmp_path, gpu_device, schedule):
     assert len(got) == len(want)
     for g, w in zip(got, want):
         torch.testing.assert_close(g, w, atol=5e-3, rtol=5e-2)
+
+
+def _pp_mem(rank, world, out_dir, schedule="gpipe"):
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import SCHEDULES, Worker, init_topology
+
+    dev = torch.device("cuda", 0)
+    topo = init_topology(dp=1, pp=world, backend="gloo", device=dev)
+    sizes = [1024] * 8 + [10, 10]
+    model = MLP(sizes, topo.stage_id, world, 4096,
+                loss="mse").materialize_device(dev)
+    opt = SGD(model.parameters(), lr=0.01)
+    ds = Dataset(4096, 256, n_samples=8192, in_dim=sizes[0],
+                 n_classes=sizes[-1], device=dev).load(0, 1)
+    w = Worker(topo, model, ds, opt)
+    torch.cuda.reset_peak_memory_stats()
+    base = torch.cuda.memory_allocated()
+    w.execute(SCHEDULES[schedule](ds.num_mubatches(), world,
+                                  topo.stage_id), 0)
+    torch.cuda.synchronize()
+    peak = torch.cuda.max_memory_allocated() - base
+    with open(os.path.join(out_dir, f"mem_{schedule}_{topo.stage_id}"),
+              "w") as f:
+        f.write(str(peak))
+    torch.distributed.destroy_process_group()
+
+
+def test_1f1b_bounds_activation_memory(tmp_path, gpu_device):
+    """PipeDream-Flush's raison d'être: stage-0 peak activation memory
+    is bounded by warmup+1 µbatches, while GPipe holds all M.  With 16
+    µbatches and 2 stages the stage-0 peak must be several times
+    smaller under 1F1B (this also guards the deferred-wgrad window —
+    unbounded deferral would silently destroy the bound)."""
+    _run(_pp_mem, 2, tmp_path, schedule="gpipe")
+    gpipe = int((tmp_path / "mem_gpipe_0").read_text())
+    _run(_pp_mem, 2, tmp_path, schedule="pipedream")
+    flush = int((tmp_path / "mem_pipedream_0").read_text())
+    assert flush * 3 < gpipe, (flush, gpipe)

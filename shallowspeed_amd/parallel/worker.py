@@ -123,13 +123,14 @@ class Worker:
             and not getattr(self, "_force_no_defer", False)
         )
         if hasattr(self.model, "set_defer_wgrad"):
-            # window = schedule's in-flight bound: GPipe (O(M) stash by
-            # design) batches all µbatches; 1F1B keeps its warmup+1
-            # memory bound and still batches within the window
-            self.model.set_defer_wgrad(
-                self._defer_active,
-                window=getattr(schedule, "max_in_flight",
-                               schedule.num_micro_batches))
+            # window = schedule's in-flight bound, floored at 4:
+            # GPipe (O(M) stash by design) batches all µbatches; 1F1B
+            # keeps its ~warmup+1 memory bound to within a small
+            # constant (≤4 retained µbatch grad/act pairs) while still
+            # amortizing kernel launches 4x
+            win = max(4, getattr(schedule, "max_in_flight",
+                                 schedule.num_micro_batches))
+            self.model.set_defer_wgrad(self._defer_active, window=win)
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)

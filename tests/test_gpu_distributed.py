@@ -126,7 +126,7 @@ def _pp_mem(rank, world, out_dir, schedule="gpipe"):
     model = MLP(sizes, topo.stage_id, world, 4096,
                 loss="mse").materialize_device(dev)
     opt = SGD(model.parameters(), lr=0.01)
-    ds = Dataset(4096, 256, n_samples=8192, in_dim=sizes[0],
+    ds = Dataset(4096, 128, n_samples=8192, in_dim=sizes[0],
                  n_classes=sizes[-1], device=dev).load(0, 1)
     w = Worker(topo, model, ds, opt)
     torch.cuda.reset_peak_memory_stats()
@@ -146,7 +146,8 @@ def test_1f1b_bounds_activation_memory(tmp_path, gpu_device):
     is bounded by warmup+1 µbatches, while GPipe holds all M.  With 16
     µbatches and 2 stages the stage-0 peak must be several times
     smaller under 1F1B (this also guards the deferred-wgrad window —
-    unbounded deferral would silently destroy the bound)."""
+    unbounded deferral would silently destroy the bound; the window
+    retains at most 4 extra µbatches)."""
     _run(_pp_mem, 2, tmp_path, schedule="gpipe")
     gpipe = int((tmp_path / "mem_gpipe_0").read_text())
     _run(_pp_mem, 2, tmp_path, schedule="pipedream")

@@ -258,6 +258,12 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
     # capture — the pinned buffer stays alive with stable content, and
     # replays re-copy it (chunk pointers are stable under the graph
     # memory pool).
+    rows = []
+    for dy, x, m in chunks:
+        assert dy.shape == dy0.shape and (m is not None) == has_mask
+        rows.append(dy.data_ptr())
+        rows.append(x.data_ptr())
+        rows.append(m.data_ptr() if m is not None else 0)
     key = (grad_w.data_ptr(), len(chunks))
     entry = _WGRAD_TABLES.get(key)
     if entry is None:
@@ -265,14 +271,15 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
                              pin_memory=True)
         dev = torch.empty(len(chunks), 3, dtype=torch.int64,
                           device=dy0.device)
-        entry = _WGRAD_TABLES[key] = (pinned, dev)
-    pinned, table = entry
-    for i, (dy, x, m) in enumerate(chunks):
-        assert dy.shape == dy0.shape and (m is not None) == has_mask
-        pinned[i, 0] = dy.data_ptr()
-        pinned[i, 1] = x.data_ptr()
-        pinned[i, 2] = m.data_ptr() if m is not None else 0
-    table.copy_(pinned, non_blocking=True)
+        entry = _WGRAD_TABLES[key] = [pinned, dev, None]
+    pinned, table, last_rows = entry
+    if rows != last_rows:
+        # chunk pointers are usually STABLE step to step (caching
+        # allocator steady state / graph pool), so the host-side table
+        # rebuild + H2D is skipped on the hot path
+        pinned.view(-1).copy_(torch.tensor(rows, dtype=torch.int64))
+        table.copy_(pinned, non_blocking=True)
+        entry[2] = rows
     ext.wgrad_tn_multi(table, len(chunks), has_mask, grad_w,
                        grad_b if grad_b is not None else torch.Tensor(),
                        dy0.shape[1], chunks[0][1].shape[1], dy0.shape[0],

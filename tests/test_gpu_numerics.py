@@ -436,8 +436,8 @@ def test_gpu_sgd_momentum_weight_decay(gpu_device):
 
 
 def test_gpu_deferred_wgrad_matches_eager(gpu_device):
-    """Pipeline schedules defer per-µbatch wgrads into ONE chunked
-    kernel per layer; grads must match the eager per-µbatch path."""
+    """Pipeline schedules defer per-µbatch wgrads into chunked
+    kernels; grads must match the eager per-µbatch path."""
     from shallowspeed_amd.data import Dataset
     from shallowspeed_amd.models import MLP, SGD
     from shallowspeed_amd.parallel import GPipeSchedule, Topology, Worker
@@ -445,52 +445,19 @@ def test_gpu_deferred_wgrad_matches_eager(gpu_device):
     def run(defer):
         model = MLP([96, 64, 48, 10], 0, 1, 512, loss="xent")
         model.materialize_device(gpu_device)
+        if not defer:
+            model.set_defer_wgrad = lambda *a, **k: None  # force eager
         opt = SGD(model.parameters(), lr=0.0)  # keep weights fixed
         ds = Dataset(512, 128, n_samples=512, in_dim=96, n_classes=10,
                      device=gpu_device).load(0, 1)
         w = Worker(Topology(device=gpu_device), model, ds, opt)
-        sched = GPipeSchedule(ds.num_mubatches(), 1, 0)
-        w.execute(sched, 0)
-        if not defer:
-            assert not w._defer_active or True
+        w.execute(GPipeSchedule(ds.num_mubatches(), 1, 0), 0)
         torch.cuda.synchronize()
-        return [p.grad.clone().cpu() for p in model.parameters()], w
+        if defer:
+            assert w._defer_active, "deferred mode should engage"
+        return [p.grad.clone().cpu() for p in model.parameters()]
 
-    got, w = run(True)
-    assert w._defer_active, "deferred mode should engage for µbatched GPU"
-    # eager reference: force-disable deferral
-    from shallowspeed_amd.data import Dataset
-    from shallowspeed_amd.models import MLP, SGD
-    from shallowspeed_amd.parallel import GPipeSchedule, Topology, Worker
-
-    model = MLP([96, 64, 48, 10], 0, 1, 512, loss="xent")
-    model.materialize_device(gpu_device)
-    opt = SGD(model.parameters(), lr=0.0)
-    ds = Dataset(512, 128, n_samples=512, in_dim=96, n_classes=10,
-                 device=gpu_device).load(0, 1)
-    w2 = Worker(Topology(device=gpu_device), model, ds, opt)
-    sched = GPipeSchedule(ds.num_mubatches(), 1, 0)
-    w2._DISPATCH = dict(w2._DISPATCH)
-    orig_exec = w2.execute
-
-    def exec_no_defer(s, b):
-        orig_exec(s, b)
-    # monkeypatch: disable after execute sets it — instead disable via
-    # model flag after worker toggles: simplest is to run execute then
-    # verify; force off by patching the worker attribute pre-dispatch
-    import shallowspeed_amd.parallel.worker as wk
-    old = wk.Worker.execute
-
-    def patched(self, schedule, batch_id):
-        r = old(self, schedule, batch_id)
-        return r
-    # direct approach: temporarily make set_defer_wgrad a no-op
-    model.set_defer_wgrad = lambda flag: None
-    for layer in model.layers:
-        if hasattr(layer, "_defer_wgrad"):
-            layer._defer_wgrad = False
-    w2.execute(sched, 0)
-    torch.cuda.synchronize()
-    want = [p.grad.clone().cpu() for p in model.parameters()]
+    got = run(True)
+    want = run(False)
     for g, r in zip(got, want):
         torch.testing.assert_close(g, r, atol=2e-3, rtol=2e-2)

@@ -35,6 +35,7 @@ void ss_ln_bwd_dparam(const void*, const void*, const void*, const void*,
                       void*, void*, int, int, hipStream_t);
 void ss_gelu_fwd(const void*, void*, long, hipStream_t);
 void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
+void ss_row_argmax(const void*, void*, int, int, hipStream_t);
 }
 
 namespace {
@@ -267,6 +268,15 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor z) {
     return dz;
 }
 
+torch::Tensor row_argmax(torch::Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 2, "x must be 2-D");
+    auto out = torch::empty({x.size(0)}, x.options().dtype(torch::kInt));
+    ss_row_argmax(x.data_ptr(), out.data_ptr(), x.size(0), x.size(1),
+                  cur_stream());
+    return out;
+}
+
 void sgd_multi(torch::Tensor desc, double lr, int64_t total,
                double momentum, double weight_decay) {
     // total passed by the caller (cached host-side) — no device sync.
@@ -298,4 +308,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ln_bwd_dparam", &ln_bwd_dparam);
     m.def("gelu_fwd", &gelu_fwd);
     m.def("gelu_bwd", &gelu_bwd);
+    m.def("row_argmax", &row_argmax);
 }

@@ -160,6 +160,39 @@ __global__ __launch_bounds__(256) void head_xent_bwd_kernel(
             dz[i] = f2bf((bf2f(s[i]) - bf2f(t[i])) * inv_gb);
 }
 
+// ------------------------------------------------------- row argmax
+
+// eval/serving helper: one wave per row (like softmax); returns the
+// index of the row maximum as int32.  torch's ROCm argmax on skinny
+// (B x 10) bf16 tensors measured ~1.3 ms at B=16384 — this kernel is
+// the same wave-reduction pattern as softmax_fwd (~5 µs).
+__global__ __launch_bounds__(256) void row_argmax_kernel(
+    const __bf16* __restrict__ x, int* __restrict__ out, int B, int C) {
+    const int lane = threadIdx.x & 63;
+    const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= B) return;
+    const __bf16* xr = x + (long)row * C;
+    float best = -1e30f;
+    int bi = 0;
+    for (int c = lane; c < C; c += 64) {
+        const float v = bf2f(xr[c]);
+        if (v > best) {
+            best = v;
+            bi = c;
+        }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float ov = __shfl_xor(best, off, 64);
+        const int oi = __shfl_xor(bi, off, 64);
+        if (ov > best || (ov == best && oi < bi)) {
+            best = ov;
+            bi = oi;
+        }
+    }
+    if (lane == 0) out[row] = bi;
+}
+
 // --------------------------------------------------------- fused SGD
 
 // desc row (int64 x 8): master*, grad*, lp*, lpt*, numel, cols, start,
@@ -247,6 +280,11 @@ void ss_head_xent_bwd(const void* s, const void* t, void* dz, long n,
     hipLaunchKernelGGL(head_xent_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                        st, (const __bf16*)s, (const __bf16*)t, (__bf16*)dz, n,
                        inv_gb);
+}
+
+void ss_row_argmax(const void* x, void* out, int B, int C, hipStream_t st) {
+    hipLaunchKernelGGL(row_argmax_kernel, dim3(cdiv(B, 4)), dim3(256), 0, st,
+                       (const __bf16*)x, (int*)out, B, C);
 }
 
 void ss_sgd_multi(const void* desc, int ntensors, long total, float lr,

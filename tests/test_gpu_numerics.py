@@ -461,3 +461,15 @@ def test_gpu_deferred_wgrad_matches_eager(gpu_device):
     want = run(False)
     for g, r in zip(got, want):
         torch.testing.assert_close(g, r, atol=2e-3, rtol=2e-2)
+
+
+def test_gpu_row_argmax(gpu_device):
+    from shallowspeed_amd.ops.functional import row_argmax
+
+    x = rand_bf16(4097, 10, device=gpu_device, scale=3.0, seed=30)
+    got = row_argmax(x)
+    want = x.float().argmax(-1)
+    torch.testing.assert_close(got.cpu(), want.cpu())
+    x2 = rand_bf16(33, 1000, device=gpu_device, seed=31)
+    torch.testing.assert_close(row_argmax(x2).cpu(),
+                               x2.float().argmax(-1).cpu())

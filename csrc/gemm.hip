@@ -44,9 +44,9 @@ __device__ __forceinline__ int kswz(int m, int k) {
 // (registers -> padded LDS tile) so the HBM latency of tile t+1 hides
 // under tile t's MFMA work (async-STAGE split, one barrier per K-step;
 // cdna_hip_programming.md Guideline 15 / T14).
-template <int TILE_ROWS, int KSTEP, bool HAS_MASK>
+template <int TILE_ROWS, int KSTEP, int NT, bool HAS_MASK>
 struct StageReg {
-    static constexpr int EL = TILE_ROWS * KSTEP / 256;  // 4..32
+    static constexpr int EL = TILE_ROWS * KSTEP / NT;  // 4..32
     __bf16 v[EL];
 
     __device__ __forceinline__ void load(const __bf16* __restrict__ src,
@@ -122,7 +122,7 @@ struct StageReg {
 
 template <int BM, int BN, int KSTEP, int WAVES_M, int WAVES_N,
           bool HAS_BIAS, bool RELU, bool HAS_MASK>
-__global__ __launch_bounds__(256) void gemm_nt_kernel(
+__global__ __launch_bounds__(WAVES_M * WAVES_N * 64) void gemm_nt_kernel(
     const __bf16* __restrict__ A,     // [M][K]
     const __bf16* __restrict__ B,     // [N][K]
     const __bf16* __restrict__ bias,  // [N]
@@ -164,8 +164,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const int lrow = lane & 15;
     const int kch = lane >> 4;  // 0..3 -> k offset kch*8
 
-    StageReg<BM, KSTEP, HAS_MASK> ra;
-    StageReg<BN, KSTEP, false> rb;
+    StageReg<BM, KSTEP, WAVES_M * WAVES_N * 64, HAS_MASK> ra;
+    StageReg<BN, KSTEP, WAVES_M * WAVES_N * 64, false> rb;
     ra.load(A, mask, M, K, m0, 0, tid);
     rb.load(B, nullptr, N, K, n0, 0, tid);
     ra.write(As[0], tid);
@@ -493,7 +493,6 @@ extern "C" {
 void ss_gemm_nt(const void* A, const void* B, const void* bias,
                 const void* mask, void* C, int M, int N, int K,
                 bool relu, hipStream_t stream) {
-    dim3 block(256);
     const bool has_bias = bias != nullptr;
     const bool has_mask = mask != nullptr;
 
@@ -504,7 +503,11 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
         constexpr bool HM = decltype(mask_tag)::value;
         constexpr int BN = BM == 128 ? 128 : 64;
         constexpr int WAVES_M = BM == 32 ? 1 : 2;
+        // 4 waves 2x2 everywhere ≥64: 8-wave 128² measured WORSE
+        // (623 vs 720 TF @8k³) without the full 8-phase schedule —
+        // exactly the guide's T3/T5 regime gate
         constexpr int WAVES_N = BM == 32 ? 4 : 2;
+        dim3 block(WAVES_M * WAVES_N * 64);
         dim3 grid(cdiv(M, BM), cdiv(N, BN));
         // Measured: KSTEP=64 is throughput-neutral at these shapes
         // (barrier savings offset by the occupancy drop 7->4

@@ -102,7 +102,6 @@ class Worker:
     # --------------------------------------------------------- execution
     def execute(self, schedule, batch_id: int):
         self._batch_id = batch_id
-        self._snapshot_inputs = getattr(schedule, "max_in_flight", 2) > 1
         # the last training stage's forward output is consumed only via
         # the head's stash (the out buffer would be dead weight: it is
         # overwritten by LoadMuBatchTarget into the grad-in pool) —
@@ -131,6 +130,14 @@ class Worker:
             win = max(4, getattr(schedule, "max_in_flight",
                                  schedule.num_micro_batches))
             self.model.set_defer_wgrad(self._defer_active, window=win)
+        # snapshot recv'd/loaded activations when layers may retain
+        # them past the buffer's reuse: >1 µbatch in flight, or wgrad
+        # deferral holding them until flush.  (Zero-copy dataset VIEWS
+        # are immutable and never need the snapshot.)
+        self._snapshot_inputs = (
+            getattr(schedule, "max_in_flight", 2) > 1
+            or self._defer_active
+        )
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
@@ -313,9 +320,10 @@ class Worker:
 
     def _forward(self, cmd):
         x = self._in_views.get(cmd.in_buffer)
+        from_view = x is not None
         if x is None:
             x = self._in_bufs[cmd.in_buffer]
-        if self.model._training and self._snapshot_inputs:
+        if self.model._training and self._snapshot_inputs and not from_view:
             # Layers stash their input per µbatch for wgrad; the input
             # buffer is SHARED across µbatches (overwritten by the next
             # Load/Recv), so snapshot it when more than one µbatch can
@@ -330,7 +338,16 @@ class Worker:
         self._out_bufs[cmd.out_buffer].copy_(y)
 
     def _backward_acc(self, cmd):
-        d = self.model.backward(self._gin_bufs[cmd.out_buffer], cmd.mubatch_id)
+        g = self._gin_bufs[cmd.out_buffer]
+        if getattr(self, "_defer_active", False) and \
+                self.topo.stage_id != self.topo.pp - 1:
+            # non-last stages hand the RECV'D grad buffer to the
+            # stage's last Linear, whose deferred wgrad would retain
+            # the reference past the buffer's next overwrite — snapshot
+            # it.  (The last stage's gin holds the TARGET, which the
+            # loss head consumes immediately.)
+            g = g.clone()
+        d = self.model.backward(g, cmd.mubatch_id)
         # stage 0 never sends input grads, so skip the staging copy
         if d is not None and self.topo.stage_id != 0:
             self._wait_buffer("gout", cmd.in_buffer)

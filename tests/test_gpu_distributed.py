@@ -73,6 +73,53 @@ def _dp_gpu(rank, world, out_dir):
     torch.distributed.destroy_process_group()
 
 
+def _pp_grads(rank, world, out_dir, schedule="gpipe"):
+    """One batch, lr=0: stage grads must match serial grads tightly —
+    catches buffer-aliasing bugs that loose weight tolerances hide."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import SCHEDULES, Worker, init_topology
+
+    dev = torch.device("cuda", 0)
+    topo = init_topology(dp=1, pp=world, backend="gloo", device=dev)
+    model = MLP(SIZES, topo.stage_id, world, GBS,
+                loss="mse").materialize_device(dev)
+    opt = SGD(model.parameters(), lr=0.0)
+    ds = Dataset(GBS, GBS // MUB, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1], device=dev).load(0, 1)
+    w = Worker(topo, model, ds, opt)
+    w.execute(SCHEDULES[schedule](ds.num_mubatches(), world,
+                                  topo.stage_id), 0)
+    torch.cuda.synchronize()
+    torch.save([p.grad.float().cpu() for p in model.parameters()],
+               os.path.join(out_dir, f"ppg_{topo.stage_id}.pt"))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.parametrize("schedule", ["gpipe", "pipedream", "naive"])
+def test_pp2_grads_match_serial_tight(tmp_path, gpu_device, schedule):
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.parallel import NaiveParallelSchedule, Topology, Worker
+
+    _run(_pp_grads, 2, tmp_path, schedule=schedule)
+    got = torch.load(tmp_path / "ppg_0.pt", weights_only=False) + \
+        torch.load(tmp_path / "ppg_1.pt", weights_only=False)
+
+    dev = torch.device("cuda", 0)
+    model = MLP(SIZES, 0, 1, GBS, loss="mse").materialize_device(dev)
+    opt = SGD(model.parameters(), lr=0.0)
+    ds = Dataset(GBS, GBS // MUB, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1], device=dev).load(0, 1)
+    w = Worker(Topology(device=dev), model, ds, opt)
+    w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), 0)
+    torch.cuda.synchronize()
+    want = [p.grad.float().cpu() for p in model.parameters()]
+    assert len(got) == len(want)
+    for g, r in zip(got, want):
+        torch.testing.assert_close(g, r, atol=2e-3, rtol=1e-2)
+
+
 def _pp_gpu(rank, world, out_dir, schedule="pipedream"):
     from shallowspeed_amd.data import Dataset
     from shallowspeed_amd.models import MLP, SGD

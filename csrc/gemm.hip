@@ -232,6 +232,133 @@ __global__ __launch_bounds__(WAVES_M * WAVES_N * 64) void gemm_nt_kernel(
     }
 }
 
+// ------------------------------------------------- gemm_nt (glds tier)
+
+// Compute-bound tier (aligned ≥128-multiple shapes, no fused mask):
+// stages tiles with gfx950 global_load_lds (direct HBM->LDS DMA, no
+// VGPR round trip; the CDNA4 guide's ladder measures this as the
+// dominant lever for the 128² structure).  LDS image is LANE-LINEAR
+// (glds writes base+lane*16), so the bank-conflict fix is an XOR
+// swizzle applied to BOTH the per-lane global SOURCE address and the
+// fragment reads (rule 21): 16-B chunk c of row r lives at physical
+// chunk c ^ ((r>>2)&3) — fragment-read banks become all-distinct.
+template <bool HAS_BIAS, bool RELU>
+__global__ __launch_bounds__(256) void gemm_nt_glds_kernel(
+    const __bf16* __restrict__ A,     // [M][K], M%128==0, K%32==0
+    const __bf16* __restrict__ B,     // [N][K], N%128==0
+    const __bf16* __restrict__ bias,  // [N]
+    __bf16* __restrict__ C,           // [M][N]
+    int M, int N, int K) {
+    constexpr int BM = 128, BN = 128, BKG = 32;
+    typedef __attribute__((address_space(1))) const unsigned int* gptr_t;
+    typedef __attribute__((address_space(3))) unsigned int* lptr_t;
+
+    __shared__ ushort As[2][BM * BKG];  // lane-linear [128][32] bf16
+    __shared__ ushort Bs[2][BN * BKG];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = wave >> 1;  // 2x2 waves, wave tile 64x64
+    const int wn = wave & 1;
+
+    // XCD-aware bijective remap (see gemm_nt_kernel)
+    const int gx = gridDim.x, gy = gridDim.y;
+    const int nwg = gx * gy;
+    const int hw = blockIdx.x + gx * blockIdx.y;
+    const int xcd = hw % 8, q8 = nwg / 8, r8 = nwg % 8;
+    const int wid = (xcd < r8 ? xcd * (q8 + 1)
+                              : r8 * (q8 + 1) + (xcd - r8) * q8) + hw / 8;
+    const int m0 = ((wid / gy) % gx) * BM;
+    const int n0 = (wid % gy) * BN;
+
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    // staging: glds is a WAVE instruction — each of the 4 waves gets
+    // its own wave-uniform LDS base (16-row span); lane l covers
+    // (row = l>>2 within the span, physical 16-B chunk p = l&3).  The
+    // source chunk is PRE-SWIZZLED: c = p ^ ((row>>2)&3), matching
+    // the fragment-read XOR (rule 21: both sides or neither).
+    const int lrow4 = lane >> 2;  // 0..15 rows within the wave span
+    const int p16 = lane & 3;     // physical chunk
+    auto stage = [&](int buf, int k0) {
+#pragma unroll
+        for (int half = 0; half < 2; ++half) {
+            const int rA = half * 64 + wave * 16 + lrow4;
+            const int cA = (p16 ^ ((rA >> 2) & 3)) * 8;
+            __builtin_amdgcn_global_load_lds(
+                (gptr_t)&A[(long)(m0 + rA) * K + k0 + cA],
+                (lptr_t)&As[buf][(half * 64 + wave * 16) * BKG], 16, 0, 0);
+        }
+#pragma unroll
+        for (int half = 0; half < 2; ++half) {
+            const int rB = half * 64 + wave * 16 + lrow4;
+            const int cB = (p16 ^ ((rB >> 2) & 3)) * 8;
+            __builtin_amdgcn_global_load_lds(
+                (gptr_t)&B[(long)(n0 + rB) * K + k0 + cB],
+                (lptr_t)&Bs[buf][(half * 64 + wave * 16) * BKG], 16, 0, 0);
+        }
+    };
+
+    const int lrow = lane & 15;
+    const int kch = lane >> 4;
+
+    stage(0, 0);
+    __syncthreads();
+
+    const int nsteps = K / BKG;
+    int cur = 0;
+    for (int t = 0; t < nsteps; ++t) {
+        if (t + 1 < nsteps) stage(cur ^ 1, (t + 1) * BKG);
+
+        bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int row = wm * 64 + i * 16 + lrow;
+            a_frag[i] = *(const bf16x8*)
+                &As[cur][row * BKG + (kch ^ ((row >> 2) & 3)) * 8];
+        }
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int row = wn * 64 + j * 16 + lrow;
+            b_frag[j] = *(const bf16x8*)
+                &Bs[cur][row * BKG + (kch ^ ((row >> 2) & 3)) * 8];
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+        // __syncthreads() drains the in-flight glds (hipcc emits the
+        // vmcnt(0) inside the barrier when LDS-DMA is outstanding)
+        __syncthreads();
+        cur ^= 1;
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int gcol = n0 + wn * 64 + j * 16 + lrow;
+            float bv = 0.f;
+            if constexpr (HAS_BIAS) bv = bf2f(bias[gcol]);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int grow = m0 + wm * 64 + i * 16 + kch * 4 + r;
+                float v = acc[i][j][r];
+                if constexpr (HAS_BIAS) v += bv;
+                if constexpr (RELU) v = v > 0.f ? v : 0.f;
+                C[(long)grow * N + gcol] = f2bf(v);
+            }
+        }
+    }
+}
+
 // ---------------------------------------------------------------- wgrad
 
 template <int BMT, int BNT, bool HAS_MASK>
@@ -541,6 +668,30 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
     else if (relu)               launch(BMT{}, F{}, T{}, F{});         \
     else                         launch(BMT{}, F{}, F{}, F{});
 
+    // compute-bound tier: aligned shapes with no fused mask go to the
+    // glds-staged kernel (direct HBM->LDS DMA)
+    if (!has_mask && N >= 128 && M % 128 == 0 && N % 128 == 0 &&
+        K % 32 == 0 && (long)(M / 128) * (N / 128) >= 512) {
+        dim3 grid(M / 128, N / 128);
+        dim3 blk(256);
+        if (has_bias && relu)
+            hipLaunchKernelGGL((gemm_nt_glds_kernel<true, true>), grid, blk,
+                               0, stream, (const __bf16*)A, (const __bf16*)B,
+                               (const __bf16*)bias, (__bf16*)C, M, N, K);
+        else if (has_bias)
+            hipLaunchKernelGGL((gemm_nt_glds_kernel<true, false>), grid, blk,
+                               0, stream, (const __bf16*)A, (const __bf16*)B,
+                               (const __bf16*)bias, (__bf16*)C, M, N, K);
+        else if (relu)
+            hipLaunchKernelGGL((gemm_nt_glds_kernel<false, true>), grid, blk,
+                               0, stream, (const __bf16*)A, (const __bf16*)B,
+                               (const __bf16*)bias, (__bf16*)C, M, N, K);
+        else
+            hipLaunchKernelGGL((gemm_nt_glds_kernel<false, false>), grid, blk,
+                               0, stream, (const __bf16*)A, (const __bf16*)B,
+                               (const __bf16*)bias, (__bf16*)C, M, N, K);
+        return;
+    }
     // 128x128 only when the grid still covers the CUs with >=2
     // blocks each (1 block/CU = 4 waves starves latency hiding)
     if (N >= 128 && (long)cdiv(M, 128) * cdiv(N, 128) >= 512) {

@@ -473,3 +473,23 @@ def test_gpu_row_argmax(gpu_device):
     x2 = rand_bf16(33, 1000, device=gpu_device, seed=31)
     torch.testing.assert_close(row_argmax(x2).cpu(),
                                x2.float().argmax(-1).cpu())
+
+
+@pytest.mark.parametrize("shape", [
+    (4096, 2048, 2048), (8192, 128, 256), (256, 128, 8192),
+])
+def test_gemm_nt_glds_tier(gpu_device, shape):
+    """Aligned unmasked shapes dispatch to the global_load_lds-staged
+    kernel; numerics vs f32 torch (+ bias/relu epilogues)."""
+    M, N, K = shape
+    e = ext()
+    a = rand_bf16(M, K, device=gpu_device, seed=40)
+    b = rand_bf16(N, K, device=gpu_device, seed=41)
+    bias = rand_bf16(N, device=gpu_device, seed=42)
+    empty = torch.Tensor()
+    c = e.gemm_nt(a, b, empty, empty, False)
+    ref = a.float() @ b.float().t()
+    torch.testing.assert_close(c.float(), ref, **tol(ref, atol=0.05))
+    c2 = e.gemm_nt(a, b, bias, empty, True)
+    ref2 = torch.clamp(ref + bias.float(), min=0)
+    torch.testing.assert_close(c2.float(), ref2, **tol(ref2, atol=0.05))

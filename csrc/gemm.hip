@@ -713,17 +713,24 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
     // big tiles only pay when both dims are wide enough that the
     // grid still covers the CUs AND operand re-reads dominate
     // (measured: at 256-wide layers the 64-config's block count wins)
-    const bool big = (Mo >= 256 && N >= 512) || (Mo >= 512 && N >= 256);
+    bool big = (Mo >= 256 && N >= 512) || (Mo >= 512 && N >= 256);
+    // debug/tuning overrides (read once)
+    static int env_big = []() {
+        const char* e = getenv("SS_WGRAD_BIG");
+        return e ? atoi(e) : -1;
+    }();
+    if (env_big >= 0) big = env_big != 0;
     const int bm = big ? 128 : 64, bn = big ? 128 : 64;
-    // pick split for (a) ~8 K-steps per block (k_per_split ≈ 512 —
-    // shorter slices pay too much prologue/atomic epilogue, longer
-    // ones tail-imbalance) and (b) at least ~512 blocks to cover the
-    // 256 CUs (measured sweet spots across the bench shapes)
+    // pick split for ~512 blocks total (2 per CU): more split
+    // duplicates the atomic-epilogue traffic (measured: 1024² best at
+    // exactly the coverage split; over-splitting by K depth cost
+    // +50%), less starves the CUs
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, bm) * cdiv(N, bn);
-        const int by_depth = cdiv(Kb, 512);
-        const int by_cover = cdiv(512, tiles);
-        split_k = by_depth > by_cover ? by_depth : by_cover;
+        split_k = cdiv(512, tiles);
+        // round DOWN to a power of two (even k-slice boundaries;
+        // measured best at 256×784)
+        while (split_k & (split_k - 1)) split_k &= split_k - 1;
         const int max_split = cdiv(Kb, 64);
         if (split_k > max_split) split_k = max_split;
     }
@@ -751,9 +758,8 @@ void ss_wgrad_tn_multi(const void* chunk_table, int nchunks, bool has_mask,
     const int bm = big ? 128 : 64, bn = big ? 128 : 64;
     if (split_k <= 0) {
         const int tiles = cdiv(Mo, bm) * cdiv(N, bn) * nchunks;
-        const int by_depth = cdiv(Kb_chunk, 512);
-        const int by_cover = cdiv(512, tiles);
-        split_k = by_depth > by_cover ? by_depth : by_cover;
+        split_k = cdiv(512, tiles);
+        while (split_k & (split_k - 1)) split_k &= split_k - 1;
         const int max_split = cdiv(Kb_chunk, 64);
         if (split_k > max_split) split_k = max_split;
     }

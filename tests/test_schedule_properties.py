@@ -1,0 +1,74 @@
+"""Property-based schedule coverage (hypothesis): any (schedule, P, M)
+combination must be deadlock-free under the Worker's comm semantics,
+conserve µbatches, and respect ordering invariants — the exhaustive
+version of the reference's hand-picked cases
+(tests/test_schedules.py:29-102 + its strengthening TODO)."""
+
+import hypothesis.strategies as st
+from hypothesis import given, settings
+
+from shallowspeed_amd.parallel.instructions import (
+    BackwardGradAcc,
+    BackwardGradAllReduce,
+    Forward,
+    OptimizerStep,
+    ZeroGrad,
+)
+from shallowspeed_amd.parallel.schedules import (
+    GPipeSchedule,
+    InferenceSchedule,
+    NaiveParallelSchedule,
+    PipeDreamFlushSchedule,
+)
+from test_schedules import flat, simulate_pipeline
+
+TRAIN = [NaiveParallelSchedule, GPipeSchedule, PipeDreamFlushSchedule]
+
+
+@given(cls=st.sampled_from(TRAIN + [InferenceSchedule]),
+       P=st.integers(1, 6), M=st.integers(1, 12))
+@settings(max_examples=120, deadline=None)
+def test_any_pipeline_progresses(cls, P, M):
+    simulate_pipeline(cls, M, P)
+
+
+@given(cls=st.sampled_from(TRAIN), P=st.integers(1, 6),
+       M=st.integers(1, 12), s=st.integers(0, 5))
+@settings(max_examples=150, deadline=None)
+def test_invariants_any_stage(cls, P, M, s):
+    if s >= P:
+        s = s % P
+    cmds = flat(cls(M, P, s))
+    # bookends
+    assert isinstance(cmds[0], ZeroGrad) and isinstance(cmds[-1], OptimizerStep)
+    # µbatch conservation, F-before-B, exactly one all-reduce backward
+    fwd = [c.mubatch_id for c in cmds if isinstance(c, Forward)]
+    bwd = [(i, c.mubatch_id) for i, c in enumerate(cmds)
+           if isinstance(c, (BackwardGradAcc, BackwardGradAllReduce))]
+    assert sorted(fwd) == list(range(M))
+    assert sorted(m for _, m in bwd) == list(range(M))
+    assert sum(isinstance(c, BackwardGradAllReduce) for c in cmds) == 1
+    pos_f = {c.mubatch_id: i for i, c in enumerate(cmds)
+             if isinstance(c, Forward)}
+    for i, m in bwd:
+        assert pos_f[m] < i
+    # the all-reduce backward is the final backward (grads complete)
+    last_bwd_idx = max(i for i, _ in bwd)
+    assert isinstance(cmds[last_bwd_idx], BackwardGradAllReduce)
+
+
+@given(P=st.integers(2, 6), M=st.integers(1, 12))
+@settings(max_examples=60, deadline=None)
+def test_1f1b_stash_bound_any_shape(P, M):
+    """Peak in-flight activations == min(M, P-s-1)+1 for every stage."""
+    for s in range(P):
+        cmds = flat(PipeDreamFlushSchedule(M, P, s))
+        inflight = peak = 0
+        for c in cmds:
+            if isinstance(c, Forward):
+                inflight += 1
+                peak = max(peak, inflight)
+            elif isinstance(c, (BackwardGradAcc, BackwardGradAllReduce)):
+                inflight -= 1
+        assert peak == min(M, P - 1 - s) + 1
+        assert inflight == 0

@@ -493,3 +493,31 @@ def test_gemm_nt_glds_tier(gpu_device, shape):
     c2 = e.gemm_nt(a, b, bias, empty, True)
     ref2 = torch.clamp(ref + bias.float(), min=0)
     torch.testing.assert_close(c2.float(), ref2, **tol(ref2, atol=0.05))
+
+
+def test_gemm_shape_fuzz(gpu_device):
+    """Odd-shape sweep through every GEMM dispatch tier (deterministic
+    pseudo-random shapes; catches bounds/edge regressions)."""
+    import random as _r
+
+    e = ext()
+    rng = _r.Random(12345)
+    empty = torch.Tensor()
+    for trial in range(24):
+        M = rng.randint(1, 400)
+        N = rng.randint(1, 300)
+        K = rng.randint(1, 300)
+        a = rand_bf16(M, K, device=gpu_device, seed=trial)
+        b = rand_bf16(N, K, device=gpu_device, seed=trial + 1000)
+        c = e.gemm_nt(a, b, empty, empty, False)
+        ref = a.float() @ b.float().t()
+        torch.testing.assert_close(c.float(), ref, **tol(ref)), (M, N, K)
+        # wgrad same shapes (Kb=M)
+        gw = torch.zeros(N, K, device=gpu_device, dtype=torch.float32)
+        gb = torch.zeros(N, device=gpu_device, dtype=torch.float32)
+        dy = rand_bf16(M, N, device=gpu_device, seed=trial + 2000)
+        x = rand_bf16(M, K, device=gpu_device, seed=trial + 3000)
+        e.wgrad_tn(dy, x, gw, gb, empty, 0)
+        rw = dy.float().t() @ x.float()
+        torch.testing.assert_close(gw, rw, **tol(rw)), (M, N, K)
+        torch.testing.assert_close(gb, dy.float().sum(0), **tol(rw))

@@ -452,7 +452,43 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const int ska = (tid * ELA) / BMT, sma = (tid * ELA) % BMT;
     const int skb = (tid * ELB) / BNT, smb = (tid * ELB) % BNT;
 
+    // glds staging (unmasked full tiles): the padded blocked image is
+    // 9 aligned 16-B pieces per 144-B subtile (8 data + 1 pad), so a
+    // per-lane SOURCE scatter performs the transpose while the LDS
+    // dest stays lane-linear; lanes landing on the pad piece load a
+    // dummy in-bounds address.  piece p -> subtile s=p/9, r=p%9;
+    // r<8: k = (s/NMT)*4 + (r>>1), m = (s%NMT)*16 + (r&1)*8.
+    typedef __attribute__((address_space(1))) const unsigned int* ggptr_t;
+    typedef __attribute__((address_space(3))) unsigned int* glptr_t;
+    auto glds_stage = [&](const __bf16* __restrict__ src, ushort* img,
+                          int nmt, int ld, int base_col, int gk0) {
+        const int total = (BKW / 4) * nmt * 9;   // pieces incl. pads
+        const int per_wave = total / 4;          // 4 waves
+        for (int i0 = 0; i0 < per_wave; i0 += 64) {
+            const int p = wave * per_wave + i0 + lane;
+            int off = 0;
+            if (lane < per_wave - i0 && (p % 9) != 8) {
+                const int sst = p / 9, rr = p % 9;
+                const int k = (sst / nmt) * 4 + (rr >> 1);
+                const int m = (sst % nmt) * 16 + (rr & 1) * 8;
+                off = (gk0 + k) * ld + base_col + m;
+            }
+            if (lane < per_wave - i0)
+                __builtin_amdgcn_global_load_lds(
+                    (ggptr_t)(src + off),
+                    (glptr_t)(img + (long)(wave * per_wave + i0) * 8), 16,
+                    0, 0);
+        }
+    };
+
     for (int k0 = kbeg; k0 < kend; k0 += BKW) {
+        const bool full_tile = !HAS_MASK && m0 + BMT <= Mo &&
+                               n0 + BNT <= N && k0 + BKW <= kend;
+        if (full_tile) {
+            glds_stage(dY, At, NMA, Mo, m0, k0);
+            glds_stage(X, Bt, NMB, N, n0, k0);
+            __syncthreads();  // drains the LDS-DMA (vmcnt inside)
+        } else {
         {
             const int gk = k0 + ska;
             if (gk < kend && m0 + BMT <= Mo) {
@@ -505,6 +541,7 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             }
         }
         __syncthreads();
+        }
 
         if (do_db) {
             constexpr int NKQ = 256 / BMT;       // threads stacked on k

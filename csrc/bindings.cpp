@@ -37,6 +37,9 @@ void ss_gelu_fwd(const void*, void*, long, hipStream_t);
 void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
 void ss_row_argmax(const void*, void*, int, int, hipStream_t);
 }
+// C++-linkage (gemm256.hip)
+bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
+                    int, bool, hipStream_t);
 
 namespace {
 
@@ -80,6 +83,24 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor bias,
     auto c = torch::empty({M, N}, a.options());
     ss_gemm_nt(a.data_ptr(), b.data_ptr(), bias_p, mask_p, c.data_ptr(), M, N,
                K, relu, cur_stream());
+    return c;
+}
+
+torch::Tensor gemm_nt_256(torch::Tensor a, torch::Tensor b,
+                          torch::Tensor bias, bool relu) {
+    check_bf16(a, "a");
+    check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "K mismatch");
+    const void* bias_p = nullptr;
+    if (has(bias)) {
+        check_bf16(bias, "bias");
+        bias_p = bias.data_ptr();
+    }
+    auto c = torch::empty({M, N}, a.options());
+    TORCH_CHECK(ss_gemm_nt_256(a.data_ptr(), b.data_ptr(), bias_p,
+                               c.data_ptr(), M, N, K, relu, cur_stream()),
+                "shape outside the 256-tile tier: ", M, "x", N, "x", K);
     return c;
 }
 
@@ -292,6 +313,8 @@ void sgd_multi(torch::Tensor desc, double lr, int64_t total,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
+    m.def("gemm_nt_256", &gemm_nt_256,
+          "256-tile 8-phase GEMM (M%256==N%256==K%128==0)");
     m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
     m.def("colsum", &colsum, "standalone column sum (bias grad)");
     m.def("wgrad_tn_multi", &wgrad_tn_multi,

@@ -185,7 +185,6 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_256_kernel(
         stageB(1, 0, t1);
         stageB(1, 1, t1);
         readB(0, 1);  // same-buffer prefetch: latency hides in the barrier
-        SS_BAR();
         SS_LGKM(4);
         mfma16(0, 0);
         SS_BAR();
@@ -193,14 +192,12 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_256_kernel(
         stageB(1, 2, t1);
         stageB(1, 3, t1);
         readA(0, 1);
-        SS_BAR();
         SS_LGKM(8);
         mfma16(0, 1);
         SS_BAR();
         // P3: q2 = (m1,n0)
         stageA(0, 0, t2);
         stageA(0, 2, t2);
-        SS_BAR();
         SS_LGKM(0);
         mfma16(1, 0);
         SS_BAR();
@@ -218,7 +215,6 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_256_kernel(
         stageB(0, 0, t2);
         stageB(0, 1, t2);
         readB(1, 1);
-        SS_BAR();
         SS_LGKM(4);
         mfma16(0, 0);
         SS_BAR();
@@ -226,14 +222,12 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_256_kernel(
         stageB(0, 2, t2);
         stageB(0, 3, t2);
         readA(1, 1);
-        SS_BAR();
         SS_LGKM(8);
         mfma16(0, 1);
         SS_BAR();
         // P7: q2
         stageA(1, 0, t3);
         stageA(1, 2, t3);
-        SS_BAR();
         SS_LGKM(0);
         mfma16(1, 0);
         SS_BAR();

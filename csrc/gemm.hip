@@ -26,6 +26,8 @@
 
 #include "common.h"
 
+#include <cstdlib>
+
 using bf16x4v = __attribute__((ext_vector_type(4))) __bf16;
 
 #define BK 32
@@ -231,6 +233,10 @@ __global__ __launch_bounds__(WAVES_M * WAVES_N * 64) void gemm_nt_kernel(
         }
     }
 }
+
+// 256-tile 8-phase tier (gemm256.hip)
+bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
+                    int, bool, hipStream_t);
 
 // ------------------------------------------------- gemm_nt (glds tier)
 
@@ -705,6 +711,19 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
     else if (relu)               launch(BMT{}, F{}, T{}, F{});         \
     else                         launch(BMT{}, F{}, F{}, F{});
 
+    // widest tier first: 256-tile 8-phase kernel (gemm256.hip) for
+    // aligned wide shapes — measured 1.3-1.6x the 128-glds tier
+    // (1027-1293 TF vs 692-811 at 4096-16384 x 4096^2, zero LDS bank
+    // conflicts; scripts/test_gemm256.py).  SS_GEMM256=0 disables.
+    if (!has_mask && N >= 512 && (long)(M / 256) * (N / 256) >= 128) {
+        static int en256 = -1;
+        if (en256 < 0) {
+            const char* e = getenv("SS_GEMM256");
+            en256 = e ? atoi(e) : 1;
+        }
+        if (en256 && ss_gemm_nt_256(A, B, bias, C, M, N, K, relu, stream))
+            return;
+    }
     // compute-bound tier: aligned shapes with no fused mask go to the
     // glds-staged kernel (direct HBM->LDS DMA)
     if (!has_mask && N >= 128 && M % 128 == 0 && N % 128 == 0 &&

@@ -37,9 +37,11 @@ void ss_gelu_fwd(const void*, void*, long, hipStream_t);
 void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
 void ss_row_argmax(const void*, void*, int, int, hipStream_t);
 }
-// C++-linkage (gemm256.hip)
+// C++-linkage (gemm256.hip / wgrad256.hip)
 bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
                     int, bool, hipStream_t);
+bool ss_wgrad_tn_256(const void*, const void*, void*, int, int, int,
+                     hipStream_t);
 
 namespace {
 
@@ -102,6 +104,18 @@ torch::Tensor gemm_nt_256(torch::Tensor a, torch::Tensor b,
                                c.data_ptr(), M, N, K, relu, cur_stream()),
                 "shape outside the 256-tile tier: ", M, "x", N, "x", K);
     return c;
+}
+
+void wgrad_tn_256(torch::Tensor dy, torch::Tensor x, torch::Tensor gw) {
+    check_bf16(dy, "dy");
+    check_bf16(x, "x");
+    check_f32(gw, "gw");
+    const int Kb = dy.size(0), Mo = dy.size(1), N = x.size(1);
+    TORCH_CHECK(x.size(0) == Kb, "batch mismatch");
+    TORCH_CHECK(gw.size(0) == Mo && gw.size(1) == N, "gw shape");
+    TORCH_CHECK(ss_wgrad_tn_256(dy.data_ptr(), x.data_ptr(), gw.data_ptr(),
+                                Mo, N, Kb, cur_stream()),
+                "shape outside the 256-tile wgrad tier");
 }
 
 void wgrad_tn(torch::Tensor dy, torch::Tensor x, torch::Tensor gw,
@@ -315,6 +329,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
     m.def("gemm_nt_256", &gemm_nt_256,
           "256-tile 8-phase GEMM (M%256==N%256==K%128==0)");
+    m.def("wgrad_tn_256", &wgrad_tn_256,
+          "256-tile 8-phase wgrad: gw += dy^T @ x (no mask/bias)");
     m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
     m.def("colsum", &colsum, "standalone column sum (bias grad)");
     m.def("wgrad_tn_multi", &wgrad_tn_multi,

@@ -234,9 +234,12 @@ __global__ __launch_bounds__(WAVES_M * WAVES_N * 64) void gemm_nt_kernel(
     }
 }
 
-// 256-tile 8-phase tier (gemm256.hip)
+// 256-tile 8-phase tiers (gemm256.hip / wgrad256.hip) — C++ linkage,
+// declared OUTSIDE the extern "C" launcher block below
 bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
                     int, bool, hipStream_t);
+bool ss_wgrad_tn_256(const void*, const void*, void*, int, int, int,
+                     hipStream_t);
 
 // ------------------------------------------------- gemm_nt (glds tier)
 
@@ -763,9 +766,28 @@ void ss_gemm_nt(const void* A, const void* B, const void* bias,
 #undef DISPATCH
 }
 
+void ss_colsum(const void*, const void*, void*, int, int, hipStream_t);
+
 void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
                  void* gb, int Mo, int N, int Kb, int split_k,
                  hipStream_t stream) {
+    // widest tier first: 8-phase 256-tile TN kernel for aligned wide
+    // unmasked shapes (bias grad via the standalone colsum kernel).
+    // SS_WGRAD256=0 disables.
+    // >= 64 output tiles: below that the split-K short loops lose to
+    // the 128-tile kernel (measured 440 vs 514 TF at 1024^2 x 16384)
+    if (!mask && split_k <= 0 && Mo >= 512 && N >= 512 &&
+        (long)(Mo / 256) * (N / 256) >= 64) {
+        static int en = -1;
+        if (en < 0) {
+            const char* e = getenv("SS_WGRAD256");
+            en = e ? atoi(e) : 1;
+        }
+        if (en && ss_wgrad_tn_256(dY, X, gW, Mo, N, Kb, stream)) {
+            if (gb) ss_colsum(dY, nullptr, gb, Kb, Mo, stream);
+            return;
+        }
+    }
     // big tiles only pay when both dims are wide enough that the
     // grid still covers the CUs AND operand re-reads dominate
     // (measured: at 256-wide layers the 64-config's block count wins)

@@ -8,10 +8,15 @@ tier and the library GEMM.  Run on a box:
 """
 
 import argparse
+import os
 import sys
 import time
 
 import torch
+
+# compare against the genuine old tiers in the perf columns
+os.environ["SS_GEMM256"] = "0"
+os.environ["SS_WGRAD256"] = "0"
 
 sys.path.insert(0, ".")
 from shallowspeed_amd.ops import load_ext  # noqa: E402
@@ -91,6 +96,57 @@ def perf(M, N, K):
           f"lib {tlib*1e6:8.1f}us {fl/tlib/1e12:7.1f}TF")
 
 
+def wg_check(Kb, Mo, N, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    dy = (torch.rand((Kb, Mo), generator=g, device=dev) * 2 - 1).bfloat16()
+    x = (torch.rand((Kb, N), generator=g, device=dev) * 2 - 1).bfloat16()
+    gw = torch.randn(Mo, N, generator=g, device=dev)  # pre-seeded: += check
+    want = gw + dy.float().t() @ x.float()
+    e.wgrad_tn_256(dy, x, gw)
+    err = ((gw - want).abs() / want.abs().clamp_min(8.0)).max().item()
+    ok = err < 0.02
+    print(f"  wgrad {Kb}x{Mo}x{N}: max_rel={err:.4f} {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def wg_race(Kb, Mo, N, runs=8, bitwise=True):
+    """bitwise=True only for split_k==1 shapes; split-K f32 atomics
+    accumulate in nondeterministic order (expected, like the 128-tile
+    split-K kernel) so those compare within tolerance."""
+    g = torch.Generator(device="cuda").manual_seed(9)
+    dy = (torch.rand((Kb, Mo), generator=g, device=dev) * 2 - 1).bfloat16()
+    x = (torch.rand((Kb, N), generator=g, device=dev) * 2 - 1).bfloat16()
+    outs = []
+    for r in range(runs):
+        gw = torch.zeros(Mo, N, device=dev)
+        _ = torch.randn(256 * (r + 1), 256, device=dev).sum()
+        e.wgrad_tn_256(dy, x, gw)
+        outs.append(gw)
+    if bitwise:
+        ok = all(torch.equal(o, outs[0]) for o in outs[1:])
+    else:
+        ok = all(torch.allclose(o, outs[0], rtol=1e-5, atol=1e-3)
+                 for o in outs[1:])
+    torch.cuda.synchronize()
+    print(f"  wgrad race {Kb}x{Mo}x{N} x{runs} "
+          f"({'bitwise' if bitwise else 'tol'}): {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def wg_perf(Kb, Mo, N):
+    dy = torch.randn(Kb, Mo, device=dev).bfloat16()
+    x = torch.randn(Kb, N, device=dev).bfloat16()
+    gw = torch.zeros(Mo, N, device=dev)
+    gb = torch.zeros(Mo, device=dev)
+    fl = 2.0 * Kb * Mo * N
+    t256 = bench_fn(lambda: e.wgrad_tn_256(dy, x, gw))
+    told = bench_fn(lambda: e.wgrad_tn(dy, x, gw, torch.Tensor(), torch.Tensor(), 0))
+    tlib = bench_fn(lambda: dy.t().float() @ x.float() if False else dy.t() @ x)
+    print(f"  wgrad {Kb}x{Mo}x{N}: 8phase {t256*1e6:8.1f}us {fl/t256/1e12:7.1f}TF | "
+          f"wgrad128 {told*1e6:8.1f}us {fl/told/1e12:7.1f}TF | "
+          f"lib(bf16 out) {tlib*1e6:8.1f}us {fl/tlib/1e12:7.1f}TF")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--perf-only", action="store_true")
@@ -110,11 +166,26 @@ def main():
         ok &= race_screen(256, 256, 512)
         ok &= race_screen(512, 512, 1024)
         ok &= race_screen(4096, 4096, 4096, runs=6)
+        print("== wgrad256 refcheck ==")
+        ok &= wg_check(128, 256, 256)
+        ok &= wg_check(384, 512, 256, seed=2)
+        ok &= wg_check(1024, 256, 768, seed=3)
+        ok &= wg_check(2048, 1024, 1024, seed=4)
+        ok &= wg_check(16384, 1024, 1024, seed=5)
+        ok &= wg_check(4096, 4096, 4096, seed=6)
+        print("== wgrad256 race ==")
+        ok &= wg_race(4096, 4096, 4096, bitwise=True)    # split_k == 1
+        ok &= wg_race(512, 512, 512, bitwise=False)      # split_k > 1
+        ok &= wg_race(16384, 1024, 1024, bitwise=False)
     print("== perf ==")
     perf(4096, 4096, 4096)
     perf(8192, 4096, 4096)
     perf(16384, 4096, 4096)
     perf(16384, 1024, 1024)
+    print("== wgrad perf ==")
+    wg_perf(16384, 4096, 4096)
+    wg_perf(4096, 4096, 4096)
+    wg_perf(16384, 1024, 1024)
     print("PASS" if ok else "FAIL")
     sys.exit(0 if ok else 1)
 

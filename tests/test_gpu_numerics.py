@@ -562,3 +562,34 @@ def test_gemm_nt_256_dispatch_consistency(gpu_device):
     for _ in range(4):
         again = e.gemm_nt(a, b, empty, empty, False)
         assert torch.equal(again, first)
+
+
+@pytest.mark.parametrize("shape", [
+    (128, 256, 256), (384, 512, 256), (2048, 1024, 1024),
+])
+def test_wgrad_tn_256_tier(gpu_device, shape):
+    """256-tile 8-phase wgrad (wgrad256.hip): accumulation semantics
+    (gw +=) vs f32 torch."""
+    Kb, Mo, N = shape
+    e = ext()
+    dy = rand_bf16(Kb, Mo, device=gpu_device, seed=70)
+    x = rand_bf16(Kb, N, device=gpu_device, seed=71)
+    gw = torch.randn(Mo, N, device=gpu_device)  # pre-seeded: += check
+    want = gw + dy.float().t() @ x.float()
+    e.wgrad_tn_256(dy, x, gw)
+    torch.testing.assert_close(gw, want, **tol(want, atol=0.05))
+
+
+def test_wgrad_tn_256_dispatch_with_bias(gpu_device):
+    """The production wgrad_tn dispatch at a qualifying wide shape
+    (256-tier + separate colsum for gb) matches f32 torch."""
+    e = ext()
+    Kb, Mo, N = 512, 2048, 2048  # 64 output tiles >= gate
+    dy = rand_bf16(Kb, Mo, device=gpu_device, seed=72)
+    x = rand_bf16(Kb, N, device=gpu_device, seed=73)
+    gw = torch.zeros(Mo, N, device=gpu_device)
+    gb = torch.zeros(Mo, device=gpu_device)
+    e.wgrad_tn(dy, x, gw, gb, torch.Tensor(), 0)
+    rw = dy.float().t() @ x.float()
+    torch.testing.assert_close(gw, rw, **tol(rw, atol=0.05))
+    torch.testing.assert_close(gb, dy.float().sum(0), **tol(rw, atol=0.05))

@@ -42,6 +42,9 @@ bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
                     int, bool, hipStream_t);
 bool ss_wgrad_tn_256(const void*, const void*, void*, int, int, int,
                      hipStream_t);
+void ss_fp8_quantize(const void*, void*, void*, int, int, hipStream_t);
+bool ss_gemm_nt_f8(const void*, const void*, const void*, const void*,
+                   const void*, void*, int, int, int, bool, hipStream_t);
 
 namespace {
 
@@ -103,6 +106,40 @@ torch::Tensor gemm_nt_256(torch::Tensor a, torch::Tensor b,
     TORCH_CHECK(ss_gemm_nt_256(a.data_ptr(), b.data_ptr(), bias_p,
                                c.data_ptr(), M, N, K, relu, cur_stream()),
                 "shape outside the 256-tile tier: ", M, "x", N, "x", K);
+    return c;
+}
+
+std::vector<torch::Tensor> fp8_quantize(torch::Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.dim() == 2 && x.size(1) % 128 == 0,
+                "x must be 2-D with K % 128 == 0");
+    const int R = x.size(0), K = x.size(1);
+    auto q = torch::empty({R, K}, x.options().dtype(torch::kUInt8));
+    auto s = torch::empty({K / 128, R, 4}, x.options().dtype(torch::kUInt8));
+    ss_fp8_quantize(x.data_ptr(), q.data_ptr(), s.data_ptr(), R, K,
+                    cur_stream());
+    return {q, s};
+}
+
+torch::Tensor gemm_nt_f8(torch::Tensor a, torch::Tensor asc, torch::Tensor b,
+                         torch::Tensor bsc, torch::Tensor bias, bool relu) {
+    TORCH_CHECK(a.scalar_type() == torch::kUInt8 &&
+                b.scalar_type() == torch::kUInt8, "a/b must be u8 (e4m3)");
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous() &&
+                asc.is_contiguous() && bsc.is_contiguous(), "contiguous");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "K mismatch");
+    const void* bias_p = nullptr;
+    if (has(bias)) {
+        check_bf16(bias, "bias");
+        bias_p = bias.data_ptr();
+    }
+    auto c = torch::empty({M, N},
+                          a.options().dtype(torch::kBFloat16));
+    TORCH_CHECK(ss_gemm_nt_f8(a.data_ptr(), asc.data_ptr(), b.data_ptr(),
+                              bsc.data_ptr(), bias_p, c.data_ptr(), M, N, K,
+                              relu, cur_stream()),
+                "shape outside the fp8 tier: ", M, "x", N, "x", K);
     return c;
 }
 
@@ -331,6 +368,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "256-tile 8-phase GEMM (M%256==N%256==K%128==0)");
     m.def("wgrad_tn_256", &wgrad_tn_256,
           "256-tile 8-phase wgrad: gw += dy^T @ x (no mask/bias)");
+    m.def("fp8_quantize", &fp8_quantize,
+          "bf16 [R,K] -> (e4m3 u8 [R,K], e8m0 scales [K/128,R,4])");
+    m.def("gemm_nt_f8", &gemm_nt_f8,
+          "MX-fp8 256-tile GEMM: C=bf16(A@B^T) (+bias)(+relu)");
     m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
     m.def("colsum", &colsum, "standalone column sum (bias grad)");
     m.def("wgrad_tn_multi", &wgrad_tn_multi,

@@ -173,6 +173,15 @@ def main():
         ok &= wg_check(2048, 1024, 1024, seed=4)
         ok &= wg_check(16384, 1024, 1024, seed=5)
         ok &= wg_check(4096, 4096, 4096, seed=6)
+        print("== fp8 tier ==")
+        ok &= f8_quant_roundtrip()
+        ok &= f8_check(256, 256, 256)
+        ok &= f8_check(512, 256, 384 + 128, seed=2)
+        ok &= f8_check(512, 512, 1024, bias_on=True, seed=3)
+        ok &= f8_check(768, 512, 2048, relu=True, seed=4)
+        ok &= f8_check(4096, 4096, 4096, bias_on=True, seed=5)
+        ok &= f8_race(512, 512, 1024)
+        ok &= f8_race(4096, 4096, 4096, runs=4)
         print("== wgrad256 race ==")
         ok &= wg_race(4096, 4096, 4096, bitwise=True)    # split_k == 1
         ok &= wg_race(512, 512, 512, bitwise=False)      # split_k > 1
@@ -182,12 +191,134 @@ def main():
     perf(8192, 4096, 4096)
     perf(16384, 4096, 4096)
     perf(16384, 1024, 1024)
+    print("== fp8 perf ==")
+    f8_perf(4096, 4096, 4096)
+    f8_perf(16384, 4096, 4096)
     print("== wgrad perf ==")
     wg_perf(16384, 4096, 4096)
     wg_perf(4096, 4096, 4096)
     wg_perf(16384, 1024, 1024)
     print("PASS" if ok else "FAIL")
     sys.exit(0 if ok else 1)
+
+
+
+
+def f8_ref(a, b, bias=None, relu=False):
+    c = a.float() @ b.float().t()
+    if bias is not None:
+        c = c + bias.float()
+    if relu:
+        c = torch.relu(c)
+    return c
+
+
+def dequant(q, s):
+    """Torch-side e4m3+E8M0 dequant — the kernel's exact oracle."""
+    R, K = q.shape
+    qf = q.to(torch.int32)
+    sgn = torch.where(qf >= 128, -1.0, 1.0)
+    qa = qf & 0x7F
+    ee = qa >> 3
+    m = (qa & 7).float()
+    mag = torch.where(ee == 0, (m / 8.0) * 2.0 ** -6,
+                      (1 + m / 8.0) * torch.pow(2.0, ee.float() - 7))
+    v = sgn * mag  # [R][K] dequant WITHOUT block scale
+    k = torch.arange(K, device=q.device)
+    g = ((k >> 6) & 1) + 2 * ((k >> 4) & 1)          # hw group of each k
+    w = k >> 7
+    sc = s.to(torch.int32)[w, :, g].t().float()       # [R][K] exponents
+    return v * torch.pow(2.0, sc - 127)
+
+
+def f8_check(M, N, K, bias_on=False, relu=False, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    a = (torch.rand((M, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    b = (torch.rand((N, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    bias = ((torch.rand((N,), generator=g, device=dev) * 2 - 1).bfloat16()
+            if bias_on else EMPTY)
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    c = e.gemm_nt_f8(qa, sa, qb, sb, bias, relu)
+    # exact oracle: f32 matmul of the dequantized operands (isolates
+    # kernel correctness from quantization noise)
+    want = dequant(qa, sa) @ dequant(qb, sb).t()
+    if bias_on:
+        want = want + bias.float()
+    if relu:
+        want = torch.relu(want)
+    err = (c.float() - want).abs()
+    rel = (err / want.abs().clamp_min(4.0)).max().item()
+    # end-to-end (vs unquantized bf16 math): loose sanity
+    full = f8_ref(a, b, bias if bias_on else None, relu)
+    e2e = ((c.float() - full).abs() /
+           full.abs().clamp_min(float(full.abs().mean()))).max().item()
+    ok = rel < 0.02
+    print(f"  fp8 {M}x{N}x{K} bias={bias_on} relu={relu}: kernel_rel="
+          f"{rel:.5f} e2e_quant_rel={e2e:.3f} {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def f8_quant_roundtrip():
+    g = torch.Generator(device="cuda").manual_seed(1)
+    x = (torch.randn((64, 256), generator=g, device=dev) * 3).bfloat16()
+    q, s = e.fp8_quantize(x)
+    # dequantize on host and compare (per-element fp8 quantization err)
+    qf = q.cpu().view(torch.uint8).float()
+    ok = True
+    import struct
+
+    def e4m3(b):
+        sgn = -1.0 if b & 0x80 else 1.0
+        ee = (b >> 3) & 0xF
+        m = b & 7
+        if ee == 0:
+            return sgn * (m / 8.0) * 2.0 ** -6
+        return sgn * (1 + m / 8.0) * 2.0 ** (ee - 7)
+
+    xe = x.float().cpu()
+    se = s.cpu()
+    err = 0.0
+    for r in range(0, 64, 17):
+        for k in range(0, 256, 37):
+            gg = ((k % 128) >> 6 & 1) + 2 * ((k >> 4) & 1)
+            sc = 2.0 ** (int(se[k // 128, r, gg]) - 127)
+            v = e4m3(int(q[r, k])) * sc
+            err = max(err, abs(v - float(xe[r, k])) / max(abs(float(xe[r, k])), 1.0))
+    ok = err < 0.08
+    print(f"  fp8 quant roundtrip: max_rel={err:.4f} {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def f8_race(M, N, K, runs=8):
+    g = torch.Generator(device="cuda").manual_seed(3)
+    a = (torch.rand((M, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    b = (torch.rand((N, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    first = e.gemm_nt_f8(qa, sa, qb, sb, EMPTY, False).clone()
+    ok = True
+    for r in range(runs - 1):
+        _ = torch.randn(512, 512, device=dev) @ torch.randn(512, 64, device=dev)
+        if not torch.equal(e.gemm_nt_f8(qa, sa, qb, sb, EMPTY, False), first):
+            ok = False
+    torch.cuda.synchronize()
+    print(f"  fp8 race {M}x{N}x{K} x{runs}: {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def f8_perf(M, N, K):
+    a = torch.randn(M, K, device=dev).bfloat16()
+    b = torch.randn(N, K, device=dev).bfloat16()
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    fl = 2.0 * M * N * K
+    t8 = bench_fn(lambda: e.gemm_nt_f8(qa, sa, qb, sb, EMPTY, False))
+    tq = bench_fn(lambda: e.fp8_quantize(a))
+    t16 = bench_fn(lambda: e.gemm_nt_256(a, b, EMPTY, False))
+    print(f"  fp8 {M}x{N}x{K}: {t8*1e6:7.1f}us {fl/t8/1e12:7.1f}TF | "
+          f"quantA {tq*1e6:6.1f}us | bf16-8phase {t16*1e6:7.1f}us "
+          f"{fl/t16/1e12:7.1f}TF")
 
 
 if __name__ == "__main__":

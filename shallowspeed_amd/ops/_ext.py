@@ -76,6 +76,7 @@ def build_ext(verbose=True):
         os.path.join(src_dir, "gemm.hip"),
         os.path.join(src_dir, "gemm256.hip"),
         os.path.join(src_dir, "wgrad256.hip"),
+        os.path.join(src_dir, "fp8.hip"),
         os.path.join(src_dir, "elementwise.hip"),
         os.path.join(src_dir, "norm.hip"),
     ]

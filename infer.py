@@ -23,6 +23,49 @@ from shallowspeed_amd.models import MLP
 from shallowspeed_amd.parallel import InferenceSchedule, Worker, init_topology
 
 
+def build_fp8_plan(model, batch):
+    """Pre-quantize qualifying Linear weights to MX-fp8 (e4m3 + E8M0
+    block scales, ops.functional.fp8_quantize); layers whose shapes
+    fall outside the fp8 tier (O%256, I%256, batch%256) stay bf16.
+    Weight quantization happens ONCE — the serving analogue of
+    weight-only deployment quantization (beyond the reference's
+    scope)."""
+    from shallowspeed_amd.models import Linear
+    from shallowspeed_amd.ops.functional import fp8_quantize
+
+    plan = []
+    for layer in model.layers:
+        if not isinstance(layer, Linear):
+            continue  # loss head: inference takes argmax of logits
+        w = layer.weight.compute()
+        bias = layer.bias.compute()
+        O, I = w.shape
+        ok = (O % 256 == 0 and I % 256 == 0 and batch % 256 == 0
+              and I >= 256 and w.is_cuda)
+        if ok:
+            wq, wsc = fp8_quantize(w)
+            plan.append(("fp8", wq, wsc, bias,
+                         layer.activation == "relu"))
+        else:
+            plan.append(("bf16", w, None, bias,
+                         layer.activation == "relu"))
+    return plan
+
+
+def fp8_forward(plan, x):
+    from shallowspeed_amd.ops.functional import (fp8_quantize, linear_fwd,
+                                                 linear_fwd_fp8)
+
+    h = x
+    for kind, w, wsc, bias, relu in plan:
+        if kind == "fp8":
+            hq, hs = fp8_quantize(h)
+            h = linear_fwd_fp8(hq, hs, w, wsc, bias, relu)
+        else:
+            h = linear_fwd(h, w, bias, relu)
+    return h
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--pp", type=int, default=1)
@@ -34,6 +77,9 @@ def main():
     ap.add_argument("--out", default=None, help="write predictions (.pt)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--backend", default=None)
+    ap.add_argument("--fp8", action="store_true",
+                    help="MX-fp8 serving path for qualifying wide layers "
+                         "(weights pre-quantized; activations per batch)")
     args = ap.parse_args()
 
     sizes = [int(s) for s in args.layer_sizes.split(",")]
@@ -58,19 +104,31 @@ def main():
     worker = Worker(topo, model, ds, None, use_dp=False)
     sched = InferenceSchedule(1, args.pp, topo.stage_id)
 
+    fp8_plan = None
+    if args.fp8:
+        assert args.pp == 1, "--fp8 serving path is single-stage"
+        fp8_plan = build_fp8_plan(model, args.batch)
+
     preds = []
     # warmup
+    if fp8_plan:
+        fp8_forward(fp8_plan, ds.x[:args.batch])
     worker.execute(sched, 0)
     if device.type == "cuda":
         torch.cuda.synchronize(device)
     t0 = time.perf_counter()
+    from shallowspeed_amd.ops.functional import row_argmax
+
     for b in range(ds.num_batches()):
+        if fp8_plan:
+            logits = fp8_forward(fp8_plan,
+                                 ds.x[b * args.batch:(b + 1) * args.batch])
+            preds.append(row_argmax(logits))
+            continue
         worker.execute(sched, b)
         if topo.stage_id == args.pp - 1:
             # HIP row-argmax, on device; one transfer at the end
             # (torch's ROCm argmax on skinny bf16 is ~1.3 ms/batch)
-            from shallowspeed_amd.ops.functional import row_argmax
-
             preds.append(row_argmax(worker._out_bufs[0]))
     if device.type == "cuda":
         torch.cuda.synchronize(device)

@@ -53,6 +53,27 @@ def linear_fwd(x, w, b=None, relu=False):
     return y
 
 
+def fp8_quantize(x):
+    """bf16 (R, K) -> (e4m3 u8 data (R, K), E8M0 scales (K/128, R, 4)).
+
+    MX block quantization matching the gfx950 scaled-MFMA's hardware
+    scale-group partition (csrc/fp8.hip; probes in
+    scripts/probe_mx*.hip).  GPU only; K % 128 == 0.
+    """
+    return _ext_for(x).fp8_quantize(x)
+
+
+def linear_fwd_fp8(xq, xs, wq, ws, b=None, relu=False):
+    """y = dequant(xq) @ dequant(wq)^T (+ b)(+ ReLU) on the MX-fp8
+    scaled-MFMA tier (~1.7x the bf16 8-phase kernel).  Opt-in serving
+    precision (beyond the reference's scope); shapes must satisfy
+    M%256 == N%256 == 0, K%256 == 0."""
+    ext = _ext_for(xq)
+    empty = torch.Tensor()
+    return ext.gemm_nt_f8(xq, xs, wq, ws, b if b is not None else empty,
+                          bool(relu))
+
+
 def linear_dgrad(dy, w, w_t=None, mask_src=None):
     """dx = (dy ⊙ 1[mask_src>0]) @ W.
 

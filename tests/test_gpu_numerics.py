@@ -593,3 +593,49 @@ def test_wgrad_tn_256_dispatch_with_bias(gpu_device):
     rw = dy.float().t() @ x.float()
     torch.testing.assert_close(gw, rw, **tol(rw, atol=0.05))
     torch.testing.assert_close(gb, dy.float().sum(0), **tol(rw, atol=0.05))
+
+
+def test_fp8_quantize_and_gemm(gpu_device):
+    """MX-fp8 tier (csrc/fp8.hip): quantizer + scaled-MFMA GEMM vs the
+    dequantized-operand f32 oracle (kernel-exact up to bf16 output
+    rounding; quantization noise is excluded by construction)."""
+    e = ext()
+
+    def dequant(q, s):
+        R, K = q.shape
+        qf = q.to(torch.int32)
+        sgn = torch.where(qf >= 128, -1.0, 1.0)
+        qa = qf & 0x7F
+        ee = qa >> 3
+        m = (qa & 7).float()
+        mag = torch.where(ee == 0, (m / 8.0) * 2.0 ** -6,
+                          (1 + m / 8.0) * torch.pow(2.0, ee.float() - 7))
+        k = torch.arange(K, device=q.device)
+        g = ((k >> 6) & 1) + 2 * ((k >> 4) & 1)
+        w = k >> 7
+        sc = s.to(torch.int32)[w, :, g].t().float()
+        return sgn * mag * torch.pow(2.0, sc - 127)
+
+    for (M, N, K, bias_on, relu) in [(256, 256, 256, False, False),
+                                     (512, 512, 512, True, True)]:
+        a = rand_bf16(M, K, device=gpu_device, seed=80)
+        b = rand_bf16(N, K, device=gpu_device, seed=81)
+        bias = (rand_bf16(N, device=gpu_device, seed=82) if bias_on
+                else torch.Tensor())
+        qa, sa = e.fp8_quantize(a)
+        qb, sb = e.fp8_quantize(b)
+        c = e.gemm_nt_f8(qa, sa, qb, sb, bias, relu)
+        want = dequant(qa, sa) @ dequant(qb, sb).t()
+        if bias_on:
+            want = want + bias.float()
+        if relu:
+            want = torch.clamp(want, min=0)
+        torch.testing.assert_close(c.float(), want, **tol(want, atol=0.05))
+        # quantization quality: e2e error vs bf16 math bounded
+        full = a.float() @ b.float().t()
+        if bias_on:
+            full += bias.float()
+        if relu:
+            full = torch.clamp(full, min=0)
+        denom = full.abs().mean().clamp_min(1.0)
+        assert ((c.float() - full).abs() / denom).mean().item() < 0.05

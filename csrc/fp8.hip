@@ -33,55 +33,41 @@ using i32x8 = __attribute__((ext_vector_type(8))) int;
 
 // ---------------------------------------------------------------- quant
 
-// Thread = one (row, 128-k window): 4 hardware-group amaxes -> E8M0
-// scales -> e4m3 data (RNE via v_cvt_pk_fp8_f32).
+// Wave = one (row, 128-k window): lane l holds k = 2l, 2l+1 (both in
+// the same hardware scale group: g = ((l>>5)&1) + 2*((l>>3)&1)); the
+// per-group amax is a 4-step shfl_xor reduction over lane bits
+// {0,1,2,4}; lanes with bits 0,1,2,4 clear write the 4 scale bytes.
 __global__ __launch_bounds__(256) void fp8_quant_kernel(
     const __bf16* __restrict__ X,  // [R][K]
     unsigned char* __restrict__ Q,       // [R][K]
     unsigned char* __restrict__ S,       // [K/128][R][4]
     int R, int K) {
-    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
     const int nwin = K / 128;
-    if (idx >= (long)R * nwin) return;
-    const int r = idx / nwin, w = idx % nwin;
-    const __bf16* src = X + (long)r * K + w * 128;
-
-    float v[128];
-    float amax[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll 4
-    for (int k = 0; k < 128; ++k) {
-        v[k] = bf2f(src[k]);
-        // hardware scale group: g = khalf + 2*parity
-        const int g = ((k >> 6) & 1) + 2 * ((k >> 4) & 1);
-        amax[g] = fmaxf(amax[g], fabsf(v[k]));
-    }
-    int e[4];
+    const long wid = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int lane = threadIdx.x & 63;
+    if (wid >= (long)R * nwin) return;  // uniform per wave
+    const int r = wid / nwin, w = wid % nwin;
+    const __bf16* src = X + (long)r * K + w * 128 + 2 * lane;
+    const float f0 = bf2f(src[0]), f1 = bf2f(src[1]);
+    float a = fmaxf(fabsf(f0), fabsf(f1));
 #pragma unroll
-    for (int g = 0; g < 4; ++g) {
-        // smallest E with amax/2^(E-127) <= 448 (e4m3 max)
-        int ex = 0;
-        if (amax[g] > 0.f) {
-            frexpf(amax[g] / 448.f, &ex);  // amax/448 = m*2^ex, m in [0.5,1)
-            ex += 127;                     // scale 2^(ex) >= amax/448
-            if (ex < 0) ex = 0;
-            if (ex > 254) ex = 254;
-        }
-        e[g] = ex;
+    for (int d : {1, 2, 4, 16}) a = fmaxf(a, __shfl_xor(a, d, 64));
+    int ex = 0;
+    if (a > 0.f) {
+        frexpf(a / 448.f, &ex);  // a/448 = m*2^ex, m in [0.5,1)
+        ex += 127;               // 2^(ex-127) >= a/448
+        if (ex < 0) ex = 0;
+        if (ex > 254) ex = 254;
+    }
+    if ((lane & 23) == 0) {  // one lane per scale group
+        const int g = ((lane >> 5) & 1) + 2 * ((lane >> 3) & 1);
         S[((long)w * R + r) * 4 + g] = (unsigned char)ex;
     }
-    unsigned char* dst = Q + (long)r * K + w * 128;
-#pragma unroll 4
-    for (int k = 0; k < 128; k += 2) {
-        const int g0 = ((k >> 6) & 1) + 2 * ((k >> 4) & 1);
-        const int g1 = (((k + 1) >> 6) & 1) + 2 * (((k + 1) >> 4) & 1);
-        const float inv0 = exp2f((float)(127 - e[g0]));
-        const float inv1 = exp2f((float)(127 - e[g1]));
-        int packed = __builtin_amdgcn_cvt_pk_fp8_f32(v[k] * inv0,
-                                                     v[k + 1] * inv1, 0,
-                                                     false);
-        dst[k] = (unsigned char)(packed & 0xff);
-        dst[k + 1] = (unsigned char)((packed >> 8) & 0xff);
-    }
+    const float inv = exp2f((float)(127 - ex));
+    const int packed =
+        __builtin_amdgcn_cvt_pk_fp8_f32(f0 * inv, f1 * inv, 0, false);
+    *(unsigned short*)(Q + (long)r * K + w * 128 + 2 * lane) =
+        (unsigned short)(packed & 0xffff);
 }
 
 // ---------------------------------------------------------------- GEMM
@@ -364,7 +350,7 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_f8_kernel(
 
 void ss_fp8_quantize(const void* X, void* Q, void* S, int R, int K,
                      hipStream_t stream) {
-    const long total = (long)R * (K / 128);
+    const long total = (long)R * (K / 128) * 64;  // one wave per window
     dim3 grid((total + 255) / 256);
     dim3 blk(256);
     hipLaunchKernelGGL(fp8_quant_kernel, grid, blk, 0, stream,

@@ -37,8 +37,8 @@ def build_fp8_plan(model, batch):
     for layer in model.layers:
         if not isinstance(layer, Linear):
             continue  # loss head: inference takes argmax of logits
-        w = layer.weight.compute()
-        bias = layer.bias.compute()
+        w = layer.weight.compute().to(torch.bfloat16)
+        bias = layer.bias.compute().to(torch.bfloat16)
         O, I = w.shape
         ok = (O % 256 == 0 and I % 256 == 0 and batch % 256 == 0
               and I >= 256 and w.is_cuda)
@@ -56,7 +56,7 @@ def fp8_forward(plan, x):
     from shallowspeed_amd.ops.functional import (fp8_quantize, linear_fwd,
                                                  linear_fwd_fp8)
 
-    h = x
+    h = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
     for kind, w, wsc, bias, relu in plan:
         if kind == "fp8":
             hq, hs = fp8_quantize(h)

@@ -639,3 +639,27 @@ def test_fp8_quantize_and_gemm(gpu_device):
             full = torch.clamp(full, min=0)
         denom = full.abs().mean().clamp_min(1.0)
         assert ((c.float() - full).abs() / denom).mean().item() < 0.05
+
+
+def test_fp8_serving_cli(gpu_device, tmp_path):
+    """infer.py --fp8 end-to-end: runs, and predictions agree with the
+    bf16 path on the overwhelming majority of samples (fp8
+    quantization may flip a few near-ties)."""
+    import json
+    import subprocess
+    import sys as _sys
+
+    outs = {}
+    for flag in ([], ["--fp8"]):
+        res = subprocess.run(
+            [_sys.executable, "infer.py", "--batch", "256", "--samples",
+             "512", "--layer-sizes", "256,512,512,256",
+             "--out", str(tmp_path / f"p{len(flag)}.pt")] + flag,
+            capture_output=True, text=True, timeout=300, check=True)
+        j = json.loads([l for l in res.stdout.splitlines()
+                        if l.startswith("{")][0])
+        assert j["inference_samples_per_sec"] > 0
+        outs[len(flag)] = torch.load(tmp_path / f"p{len(flag)}.pt",
+                                     weights_only=False)
+    agree = (outs[0] == outs[1]).float().mean().item()
+    assert agree > 0.97, f"fp8 predictions diverge: agree={agree}"

@@ -45,6 +45,9 @@ bool ss_wgrad_tn_256(const void*, const void*, void*, int, int, int,
 void ss_fp8_quantize(const void*, void*, void*, int, int, hipStream_t);
 bool ss_gemm_nt_f8(const void*, const void*, const void*, const void*,
                    const void*, void*, int, int, int, bool, hipStream_t);
+bool ss_gemm_nt_f8_q(const void*, const void*, const void*, const void*,
+                     const void*, void*, void*, int, int, int, bool,
+                     hipStream_t);
 
 namespace {
 
@@ -141,6 +144,27 @@ torch::Tensor gemm_nt_f8(torch::Tensor a, torch::Tensor asc, torch::Tensor b,
                               relu, cur_stream()),
                 "shape outside the fp8 tier: ", M, "x", N, "x", K);
     return c;
+}
+
+std::vector<torch::Tensor> gemm_nt_f8_q(torch::Tensor a, torch::Tensor asc,
+                                         torch::Tensor b, torch::Tensor bsc,
+                                         torch::Tensor bias, bool relu) {
+    TORCH_CHECK(a.scalar_type() == torch::kUInt8 &&
+                b.scalar_type() == torch::kUInt8, "a/b must be u8 (e4m3)");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "K mismatch");
+    const void* bias_p = nullptr;
+    if (has(bias)) {
+        check_bf16(bias, "bias");
+        bias_p = bias.data_ptr();
+    }
+    auto cq = torch::empty({M, N}, a.options());
+    auto cs = torch::empty({N / 128, M, 4}, a.options());
+    TORCH_CHECK(ss_gemm_nt_f8_q(a.data_ptr(), asc.data_ptr(), b.data_ptr(),
+                                bsc.data_ptr(), bias_p, cq.data_ptr(),
+                                cs.data_ptr(), M, N, K, relu, cur_stream()),
+                "shape outside the fp8 tier: ", M, "x", N, "x", K);
+    return {cq, cs};
 }
 
 void wgrad_tn_256(torch::Tensor dy, torch::Tensor x, torch::Tensor gw) {
@@ -372,6 +396,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "bf16 [R,K] -> (e4m3 u8 [R,K], e8m0 scales [K/128,R,4])");
     m.def("gemm_nt_f8", &gemm_nt_f8,
           "MX-fp8 256-tile GEMM: C=bf16(A@B^T) (+bias)(+relu)");
+    m.def("gemm_nt_f8_q", &gemm_nt_f8_q,
+          "MX-fp8 GEMM with FUSED output quantization -> (e4m3, scales)");
     m.def("wgrad_tn", &wgrad_tn, "gW += (dy⊙mask)^T @ x; gb += colsum (fused)");
     m.def("colsum", &colsum, "standalone column sum (bias grad)");
     m.def("wgrad_tn_multi", &wgrad_tn_multi,

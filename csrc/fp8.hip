@@ -72,14 +72,16 @@ __global__ __launch_bounds__(256) void fp8_quant_kernel(
 
 // ---------------------------------------------------------------- GEMM
 
-template <bool HAS_BIAS, bool RELU>
+template <bool HAS_BIAS, bool RELU, bool EMIT_Q>
 __global__ __launch_bounds__(512, 1) void gemm_nt_f8_kernel(
     const unsigned char* __restrict__ A,    // [M][K] e4m3
     const unsigned char* __restrict__ Asc,  // [K/128][M][4] e8m0
     const unsigned char* __restrict__ B,    // [N][K]
     const unsigned char* __restrict__ Bsc,  // [K/128][N][4]
     const __bf16* __restrict__ bias,        // [N]
-    __bf16* __restrict__ C,                 // [M][N]
+    __bf16* __restrict__ C,                 // [M][N] bf16 (EMIT_Q==0)
+    unsigned char* __restrict__ Cq,         // [M][N] e4m3 (EMIT_Q==1)
+    unsigned char* __restrict__ Cs,         // [N/128][M][4]
     int M, int N, int K) {
     // [op][buf][256 rows × 128 k data, then 256×4 scale rows] u8 —
     // ONE LDS object (a second __shared__ written by glds makes the
@@ -327,20 +329,73 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_f8_kernel(
     SS_VMCNT(0);
 
     // ---- epilogue -------------------------------------------------
+    // EMIT_Q: the output is quantized IN the epilogue (fused — no
+    // separate pass, no bf16 round trip): the wave's 64 columns fall
+    // in ONE 128-col scale window (khalf = wn&1) and parity = J&1,
+    // so each (row, parity) MX group amax is a 4-step shfl_xor
+    // reduction over the 16 column lanes — no LDS, no cross-wave
+    // sync.  Scale/data layouts match this kernel's own INPUT format,
+    // so serving chains stay fp8-resident end to end.
 #pragma unroll
     for (int I = 0; I < 8; ++I) {
+        float bv[4];
 #pragma unroll
         for (int J = 0; J < 4; ++J) {
-            const int gcol = n0 + wn * 64 + J * 16 + lrow;
-            float bv = 0.f;
-            if constexpr (HAS_BIAS) bv = bf2f(bias[gcol]);
+            bv[J] = 0.f;
+            if constexpr (HAS_BIAS)
+                bv[J] = bf2f(bias[n0 + wn * 64 + J * 16 + lrow]);
+        }
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int grow = m0 + wm * 128 + I * 16 + kch * 4 + r;
-                float v = acc[I][J][r];
-                if constexpr (HAS_BIAS) v += bv;
-                if constexpr (RELU) v = v > 0.f ? v : 0.f;
-                C[(long)grow * N + gcol] = f2bf(v);
+        for (int r = 0; r < 4; ++r) {
+            const int grow = m0 + wm * 128 + I * 16 + kch * 4 + r;
+            float v[4];
+#pragma unroll
+            for (int J = 0; J < 4; ++J) {
+                v[J] = acc[I][J][r];
+                if constexpr (HAS_BIAS) v[J] += bv[J];
+                if constexpr (RELU) v[J] = v[J] > 0.f ? v[J] : 0.f;
+            }
+            if constexpr (!EMIT_Q) {
+#pragma unroll
+                for (int J = 0; J < 4; ++J)
+                    C[(long)grow * N + n0 + wn * 64 + J * 16 + lrow] =
+                        f2bf(v[J]);
+            } else {
+                // group amax per parity (J&1) over J-pair + 16 lanes
+                float m0p = fmaxf(fabsf(v[0]), fabsf(v[2]));
+                float m1p = fmaxf(fabsf(v[1]), fabsf(v[3]));
+#pragma unroll
+                for (int d : {1, 2, 4, 8}) {
+                    m0p = fmaxf(m0p, __shfl_xor(m0p, d, 64));
+                    m1p = fmaxf(m1p, __shfl_xor(m1p, d, 64));
+                }
+                int e0 = 0, e1 = 0;
+                if (m0p > 0.f) {
+                    frexpf(m0p / 448.f, &e0);
+                    e0 = min(max(e0 + 127, 0), 254);
+                }
+                if (m1p > 0.f) {
+                    frexpf(m1p / 448.f, &e1);
+                    e1 = min(max(e1 + 127, 0), 254);
+                }
+                const int w = (n0 + wn * 64) >> 7;
+                if (lrow == 0) {
+                    Cs[((long)w * M + grow) * 4 + ((wn & 1) + 0)] =
+                        (unsigned char)e0;
+                    Cs[((long)w * M + grow) * 4 + ((wn & 1) + 2)] =
+                        (unsigned char)e1;
+                }
+                const float i0 = exp2f((float)(127 - e0));
+                const float i1 = exp2f((float)(127 - e1));
+                const int p02 = __builtin_amdgcn_cvt_pk_fp8_f32(
+                    v[0] * i0, v[2] * i0, 0, false);
+                const int p13 = __builtin_amdgcn_cvt_pk_fp8_f32(
+                    v[1] * i1, v[3] * i1, 0, false);
+                const long rb = (long)grow * N + n0 + wn * 64 + lrow;
+                Cq[rb] = (unsigned char)(p02 & 0xff);
+                Cq[rb + 32] = (unsigned char)((p02 >> 8) & 0xff);
+                Cq[rb + 16] = (unsigned char)(p13 & 0xff);
+                Cq[rb + 48] = (unsigned char)((p13 >> 8) & 0xff);
             }
         }
     }
@@ -366,33 +421,41 @@ bool ss_gemm_nt_f8(const void* A, const void* Asc, const void* B,
     dim3 grid(M / 256, N / 256);
     dim3 blk(512);
     const bool hb = bias != nullptr;
-    if (hb && relu)
-        hipLaunchKernelGGL((gemm_nt_f8_kernel<true, true>), grid, blk, 0,
-                           stream, (const unsigned char*)A,
-                           (const unsigned char*)Asc,
-                           (const unsigned char*)B,
-                           (const unsigned char*)Bsc, (const __bf16*)bias,
-                           (__bf16*)C, M, N, K);
-    else if (hb)
-        hipLaunchKernelGGL((gemm_nt_f8_kernel<true, false>), grid, blk, 0,
-                           stream, (const unsigned char*)A,
-                           (const unsigned char*)Asc,
-                           (const unsigned char*)B,
-                           (const unsigned char*)Bsc, (const __bf16*)bias,
-                           (__bf16*)C, M, N, K);
-    else if (relu)
-        hipLaunchKernelGGL((gemm_nt_f8_kernel<false, true>), grid, blk, 0,
-                           stream, (const unsigned char*)A,
-                           (const unsigned char*)Asc,
-                           (const unsigned char*)B,
-                           (const unsigned char*)Bsc, nullptr, (__bf16*)C,
-                           M, N, K);
-    else
-        hipLaunchKernelGGL((gemm_nt_f8_kernel<false, false>), grid, blk, 0,
-                           stream, (const unsigned char*)A,
-                           (const unsigned char*)Asc,
-                           (const unsigned char*)B,
-                           (const unsigned char*)Bsc, nullptr, (__bf16*)C,
-                           M, N, K);
+#define F8LAUNCH(HB, RL)                                                     \
+    hipLaunchKernelGGL((gemm_nt_f8_kernel<HB, RL, false>), grid, blk, 0,     \
+                       stream, (const unsigned char*)A,                      \
+                       (const unsigned char*)Asc, (const unsigned char*)B,   \
+                       (const unsigned char*)Bsc, (const __bf16*)bias,       \
+                       (__bf16*)C, nullptr, nullptr, M, N, K)
+    if (hb && relu) F8LAUNCH(true, true);
+    else if (hb) F8LAUNCH(true, false);
+    else if (relu) F8LAUNCH(false, true);
+    else F8LAUNCH(false, false);
+#undef F8LAUNCH
+    return true;
+}
+
+// Fused-quant output: C is emitted as (e4m3 data, E8M0 scales) in the
+// same layout this kernel consumes — fp8-resident serving chains.
+bool ss_gemm_nt_f8_q(const void* A, const void* Asc, const void* B,
+                     const void* Bsc, const void* bias, void* Cq, void* Cs,
+                     int M, int N, int K, bool relu, hipStream_t stream) {
+    if (M % 256 || N % 256 || K % 256 || K < 256) return false;
+    if ((long)M * K >= (1L << 31) || (long)N * K >= (1L << 31)) return false;
+    dim3 grid(M / 256, N / 256);
+    dim3 blk(512);
+    const bool hb = bias != nullptr;
+#define F8QLAUNCH(HB, RL)                                                    \
+    hipLaunchKernelGGL((gemm_nt_f8_kernel<HB, RL, true>), grid, blk, 0,      \
+                       stream, (const unsigned char*)A,                      \
+                       (const unsigned char*)Asc, (const unsigned char*)B,   \
+                       (const unsigned char*)Bsc, (const __bf16*)bias,       \
+                       nullptr, (unsigned char*)Cq, (unsigned char*)Cs, M,   \
+                       N, K)
+    if (hb && relu) F8QLAUNCH(true, true);
+    else if (hb) F8QLAUNCH(true, false);
+    else if (relu) F8QLAUNCH(false, true);
+    else F8QLAUNCH(false, false);
+#undef F8QLAUNCH
     return true;
 }

@@ -53,15 +53,28 @@ def build_fp8_plan(model, batch):
 
 
 def fp8_forward(plan, x):
+    """fp8-RESIDENT chain: a layer whose successor is also fp8 emits
+    its output pre-quantized from the GEMM epilogue (gemm_nt_f8_q) —
+    activations never round-trip through bf16 between fp8 layers."""
+    from shallowspeed_amd.ops import load_ext
     from shallowspeed_amd.ops.functional import (fp8_quantize, linear_fwd,
                                                  linear_fwd_fp8)
 
+    e = load_ext(required=True)
     h = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
-    for kind, w, wsc, bias, relu in plan:
+    hq = hs = None  # fp8-resident activation, when set
+    for li, (kind, w, wsc, bias, relu) in enumerate(plan):
+        nxt_fp8 = li + 1 < len(plan) and plan[li + 1][0] == "fp8"
         if kind == "fp8":
-            hq, hs = fp8_quantize(h)
-            h = linear_fwd_fp8(hq, hs, w, wsc, bias, relu)
+            if hq is None:
+                hq, hs = fp8_quantize(h)
+            if nxt_fp8:
+                hq, hs = e.gemm_nt_f8_q(hq, hs, w, wsc, bias, relu)
+            else:
+                h = linear_fwd_fp8(hq, hs, w, wsc, bias, relu)
+                hq = hs = None
         else:
+            assert hq is None  # nxt_fp8 guarantees bf16 h here
             h = linear_fwd(h, w, bias, relu)
     return h
 

@@ -180,6 +180,9 @@ def main():
         ok &= f8_check(512, 512, 1024, bias_on=True, seed=3)
         ok &= f8_check(768, 512, 2048, relu=True, seed=4)
         ok &= f8_check(4096, 4096, 4096, bias_on=True, seed=5)
+        ok &= f8q_check(256, 256, 256)
+        ok &= f8q_check(512, 512, 1024, relu=True, seed=2)
+        ok &= f8q_check(4096, 4096, 4096, seed=3)
         ok &= f8_race(512, 512, 1024)
         ok &= f8_race(4096, 4096, 4096, runs=4)
         print("== wgrad256 race ==")
@@ -194,6 +197,7 @@ def main():
     print("== fp8 perf ==")
     f8_perf(4096, 4096, 4096)
     f8_perf(16384, 4096, 4096)
+    f8q_perf(16384, 4096, 4096)
     print("== wgrad perf ==")
     wg_perf(16384, 4096, 4096)
     wg_perf(4096, 4096, 4096)
@@ -288,6 +292,37 @@ def f8_quant_roundtrip():
     ok = err < 0.08
     print(f"  fp8 quant roundtrip: max_rel={err:.4f} {'OK' if ok else 'FAIL'}")
     return ok
+
+
+def f8q_check(M, N, K, relu=False, seed=0):
+    """Fused-quant output: dequant(q,s) must match the bf16-out path
+    within one e4m3 quantization step."""
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    a = (torch.rand((M, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    b = (torch.rand((N, K), generator=g, device=dev) * 2 - 1).bfloat16()
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    c16 = e.gemm_nt_f8(qa, sa, qb, sb, EMPTY, relu).float()
+    cq, cs = e.gemm_nt_f8_q(qa, sa, qb, sb, EMPTY, relu)
+    cd = dequant(cq, cs)
+    rel = ((cd - c16).abs() / c16.abs().clamp_min(4.0)).max().item()
+    ok = rel < 0.07  # one e4m3 ulp
+    print(f"  f8q fused-out {M}x{N}x{K} relu={relu}: rel={rel:.4f} "
+          f"{'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def f8q_perf(M, N, K):
+    a = torch.randn(M, K, device=dev).bfloat16()
+    b = torch.randn(N, K, device=dev).bfloat16()
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    fl = 2.0 * M * N * K
+    tq = bench_fn(lambda: e.gemm_nt_f8_q(qa, sa, qb, sb, EMPTY, True))
+    t16 = bench_fn(lambda: e.gemm_nt_f8(qa, sa, qb, sb, EMPTY, True))
+    print(f"  f8q {M}x{N}x{K}: fused-q-out {tq*1e6:7.1f}us "
+          f"{fl/tq/1e12:7.1f}TF | bf16-out {t16*1e6:7.1f}us "
+          f"{fl/t16/1e12:7.1f}TF")
 
 
 def f8_race(M, N, K, runs=8):

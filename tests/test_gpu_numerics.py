@@ -667,3 +667,31 @@ def test_fp8_serving_cli(gpu_device, tmp_path):
     # fp8 path agrees at chance (1/256).  Trained-model margins give
     # far higher agreement (infer --fp8 A/B in profiles/).
     assert agree > 0.75, f"fp8 predictions diverge: agree={agree}"
+
+
+def test_fp8_fused_quant_output(gpu_device):
+    """gemm_nt_f8_q (EMIT_Q epilogue): dequantized fused output equals
+    the bf16-out path within one e4m3 ulp."""
+    e = ext()
+    M = N = K = 256
+    a = rand_bf16(M, K, device=gpu_device, seed=90)
+    b = rand_bf16(N, K, device=gpu_device, seed=91)
+    qa, sa = e.fp8_quantize(a)
+    qb, sb = e.fp8_quantize(b)
+    c16 = e.gemm_nt_f8(qa, sa, qb, sb, torch.Tensor(), False).float()
+    cq, cs = e.gemm_nt_f8_q(qa, sa, qb, sb, torch.Tensor(), False)
+    # dequant (same as test_fp8_quantize_and_gemm)
+    qf = cq.to(torch.int32)
+    sgn = torch.where(qf >= 128, -1.0, 1.0)
+    qq = qf & 0x7F
+    ee = qq >> 3
+    m = (qq & 7).float()
+    mag = torch.where(ee == 0, (m / 8.0) * 2.0 ** -6,
+                      (1 + m / 8.0) * torch.pow(2.0, ee.float() - 7))
+    k = torch.arange(N, device=gpu_device)
+    g = ((k >> 6) & 1) + 2 * ((k >> 4) & 1)
+    w = k >> 7
+    sc = cs.to(torch.int32)[w, :, g].t().float()
+    cd = sgn * mag * torch.pow(2.0, sc - 127)
+    rel = ((cd - c16).abs() / c16.abs().clamp_min(4.0)).max().item()
+    assert rel < 0.07, rel

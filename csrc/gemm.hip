@@ -401,8 +401,8 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     constexpr int FM = BMT / 32;   // 16x16 frags per wave (2x2 waves)
     constexpr int FN = BNT / 32;
 
-    __shared__ ushort At[BKW / 4 * NMA * 72];  // dY^T tile, blocked+padded
-    __shared__ ushort Bt[BKW / 4 * NMB * 72];  // X^T tile, blocked+padded
+    __shared__ ushort At[NMA * 16 * 64];  // dY^T tile, blocked+bit-swapped
+    __shared__ ushort Bt[NMB * 16 * 64];  // X^T tile, blocked+bit-swapped
     __shared__ float dbs[256 / BMT > 1 ? 256 / BMT : 2][BMT];
 
     const int tid = threadIdx.x;
@@ -445,13 +445,20 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     const bool do_db = (gb != nullptr) && (bidy == 0);
     float db_part = 0.f;
 
-    // subtiles padded 64->72 ushorts (144 B): consecutive k-subtiles no
-    // longer alias the same LDS bank row for the tr-read lane groups
-    auto baddrA = [](int k, int m) {
-        return ((k >> 2) * NMA + (m >> 4)) * 72 + (k & 3) * 16 + (m & 15);
+    // mblk-major image with the ksub bits 0,1 SWAPPED in the subtile
+    // position (wgrad256.hip's placement): the four tr-read lane
+    // groups (k-stride 2) then land on {row even|odd} x {phase 0|1}
+    // of the 256-B bank rows — the optimal 2-clk pattern with NO
+    // padding (the old 144-B-padded image left a 4-deep overlap on
+    // banks 24-31).
+    auto swsub = [](int ksub) {
+        return (ksub & ~3) | ((ksub & 1) << 1) | ((ksub >> 1) & 1);
     };
-    auto baddrB = [](int k, int m) {
-        return ((k >> 2) * NMB + (m >> 4)) * 72 + (k & 3) * 16 + (m & 15);
+    auto baddrA = [swsub](int k, int m) {
+        return ((m >> 4) * 16 + swsub(k >> 2)) * 64 + (k & 3) * 16 + (m & 15);
+    };
+    auto baddrB = [swsub](int k, int m) {
+        return ((m >> 4) * 16 + swsub(k >> 2)) * 64 + (k & 3) * 16 + (m & 15);
     };
 
     // staging: thread t covers ELA contiguous elems of the [BKW][BMT]
@@ -469,24 +476,25 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
     // r<8: k = (s/NMT)*4 + (r>>1), m = (s%NMT)*16 + (r&1)*8.
     typedef __attribute__((address_space(1))) const unsigned int* ggptr_t;
     typedef __attribute__((address_space(3))) unsigned int* glptr_t;
+    // unpadded image: subtile = exactly 8 aligned 16-B pieces, so the
+    // per-lane source scatter (the transpose) has NO pad lanes; piece
+    // q -> position p = q/8 (mblk = p/16, ksub = swsub(p&15)), r = q%8
+    // -> global (k = ksub*4 + (r>>1), m = mblk*16 + (r&1)*8).
     auto glds_stage = [&](const __bf16* __restrict__ src, ushort* img,
                           int nmt, int ld, int base_col, int gk0) {
-        const int total = (BKW / 4) * nmt * 9;   // pieces incl. pads
+        const int total = nmt * 16 * 8;          // pieces
         const int per_wave = total / 4;          // 4 waves
         for (int i0 = 0; i0 < per_wave; i0 += 64) {
-            const int p = wave * per_wave + i0 + lane;
-            int off = 0;
-            if (lane < per_wave - i0 && (p % 9) != 8) {
-                const int sst = p / 9, rr = p % 9;
-                const int k = (sst / nmt) * 4 + (rr >> 1);
-                const int m = (sst % nmt) * 16 + (rr & 1) * 8;
-                off = (gk0 + k) * ld + base_col + m;
-            }
-            if (lane < per_wave - i0)
-                __builtin_amdgcn_global_load_lds(
-                    (ggptr_t)(src + off),
-                    (glptr_t)(img + (long)(wave * per_wave + i0) * 8), 16,
-                    0, 0);
+            const int q = wave * per_wave + i0 + lane;
+            const int p = q >> 3, rr = q & 7;
+            const int ksub = ((p & 15) & ~3) | (((p & 15) & 1) << 1) |
+                             (((p & 15) >> 1) & 1);
+            const int k = ksub * 4 + (rr >> 1);
+            const int m = (p >> 4) * 16 + (rr & 1) * 8;
+            __builtin_amdgcn_global_load_lds(
+                (ggptr_t)(src + (gk0 + k) * ld + base_col + m),
+                (glptr_t)(img + (long)(wave * per_wave + i0) * 8), 16, 0,
+                0);
         }
     };
 
@@ -574,9 +582,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int i = 0; i < FM; ++i) {
                 const int mblk = wm * FM + i;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[(g2 * NMA + mblk) * 72 + lcol4]);
+                    (lds_v4p)&At[(mblk * 16 + swsub(g2)) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&At[((g2 + 1) * NMA + mblk) * 72 + lcol4]);
+                    (lds_v4p)&At[(mblk * 16 + swsub(g2 + 1)) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     a_frag[i][e] = lo[e];
@@ -587,9 +595,9 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
             for (int j = 0; j < FN; ++j) {
                 const int nblk = wn * FN + j;
                 bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[(g2 * NMB + nblk) * 72 + lcol4]);
+                    (lds_v4p)&Bt[(nblk * 16 + swsub(g2)) * 64 + lcol4]);
                 bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                    (lds_v4p)&Bt[((g2 + 1) * NMB + nblk) * 72 + lcol4]);
+                    (lds_v4p)&Bt[(nblk * 16 + swsub(g2 + 1)) * 64 + lcol4]);
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     b_frag[j][e] = lo[e];

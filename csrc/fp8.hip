@@ -10,10 +10,15 @@
 //     quantizer's amax groups match that partition exactly.
 //   * gemm_nt_f8_kernel: the gemm256.hip 8-phase skeleton at fp8:
 //     same 256×256 tile, K-tile = 128 elems (= the same 128 B per
-//     row, so the staging schedule, slot swizzle and vmcnt(4)
-//     certification carry over verbatim), MFMA = one scaled
-//     16x16x128 op per fragment (2× the bf16 rate), scale rows
-//     staged by one extra glds per operand per tile at P2/P6.
+//     row, so the staging schedule and slot swizzle carry over),
+//     same-phase fragment reads (lgkmcnt(0) before the MFMA burst;
+//     the SIMD's partner wave hides the wait), MFMA = one scaled
+//     16x16x128 op per fragment (2× the bf16 rate).  Scale rows are
+//     staged by one extra glds per operand per tile at P4/P8 (after
+//     their last reader's phase-end barrier), so the certification
+//     checkpoints are vmcnt(6).  With EMIT_Q the epilogue emits the
+//     output pre-quantized (per-row MX groups via shfl_xor over the
+//     16 column lanes) — fp8-RESIDENT serving chains.
 //
 // This is the reference-beyond parity item (ROADMAP #6): an opt-in
 // serving/inference precision mode; the bench contract stays bf16.
@@ -248,7 +253,7 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_f8_kernel(
     // ---- main loop: same-phase reads (guide-style lgkmcnt(0) before
     // MFMA — the SIMD partner wave covers the read wait); staging
     // schedule and vmcnt(6) certification as derived for gemm256 +
-    // scale rows at P2/P6.  Reads at P1/P5 touch the buffer certified
+    // scale rows at P4/P8.  Reads at P1/P5 touch the buffer certified
     // by the PREVIOUS vmcnt barrier.
     const int niter = nsteps / 2;
     for (int it = 0; it < niter; ++it) {

@@ -7,18 +7,21 @@
 // This is the deep-pipeline structure the 128²/2-barrier glds kernel
 // (gemm.hip) cannot reach: 8 waves (2M×4N, wave tile 128×64,
 // acc[8][4] = 128 VGPRs), BK=64, 2 K-tiles per unrolled iteration,
-// 8 phases of {glds staging | counted-s_waitcnt | barrier | ds_read
-// prefetch | 16 MFMA under setprio(1)}.  LDS = 128 KiB: 2 buffers ×
+// 8 phases of {2×glds | pre-barrier ds_read prefetch | counted
+// lgkmcnt | 16 MFMA under setprio(1) | barrier}, with an extra
+// vmcnt(4)+barrier certification point at P4/P8 only (barriers
+// elsewhere are thinned: the phase-end barrier alone carries the
+// freeing proof).  LDS = 128 KiB: 2 buffers ×
 // (A 256×64 + B 256×64) bf16 — legal as a single static allocation on
 // gfx950 (160 KiB/CU).
 //
 // Correctness is by a statically-derived schedule (no dynamic sync):
 //   * staging granule = 64 rows × 128 B; each wave stages an 8-row
 //     slice per glds (64 lanes × 16 B, lane-linear LDS).
-//   * per-phase glds counts {P1:4, P3:2, P4:2, P5:4, P7:2, P8:2} per
-//     wave; `s_waitcnt vmcnt(4)` at P4/P8 certifies every granule
-//     ≥3 phases before its first ds_read (derivation in the phase
-//     comments below).
+//   * uniform 2 glds per wave per phase ({P1,P2: next tile's B into
+//     the other buffer | P3,P4: A | P5,P6: B | P7,P8: next A});
+//     `s_waitcnt vmcnt(4)` at P4/P8 certifies every granule ≥2-3
+//     phases before its first ds_read (ledger in the loop comments).
 //   * fragment ds_reads run ONE phase ahead of their MFMA consumer
 //     with counted lgkmcnt (never 0 in the loop), so a K-tile's LDS
 //     is fully read two phases into its 4-phase window — freeing its
@@ -165,8 +168,9 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_256_kernel(
     readB(0, 0);
 
     // ---- main loop: 2 K-tiles (buf0 then buf1) per iteration ------
-    // Phase shape: [glds][vmcnt@P4,P8][barrier][ds_read next-phase
-    // frags][counted lgkm][16 MFMA prio1][barrier].  Staged-granule
+    // Phase shape: [2 glds][same-buffer ds_read prefetch][counted
+    // lgkm][16 MFMA prio1][end barrier]; P4/P8 insert vmcnt(4) + a
+    // barrier before their (cross-buffer) reads.  Staged-granule
     // certification ledger (vmcnt(4) = all but last 2 phases' glds):
     //   tile 2i+1.B  issued P1        first read P4   certified P4 ✓
     //   tile 2i+2.A  issued P3,P4     first read P8   certified P8 ✓

@@ -18,8 +18,9 @@
 // Correctness is by a statically-derived schedule (no dynamic sync):
 //   * staging granule = 64 rows × 128 B; each wave stages an 8-row
 //     slice per glds (64 lanes × 16 B, lane-linear LDS).
-//   * uniform 2 glds per wave per phase ({P1,P2: next tile's B into
-//     the other buffer | P3,P4: A | P5,P6: B | P7,P8: next A});
+//   * uniform 2 glds per wave per phase (P1,P2: this iteration's
+//     second tile's B | P3,P4: next tile's A | P5,P6: its B | P7,P8:
+//     the tile after's A);
 //     `s_waitcnt vmcnt(4)` at P4/P8 certifies every granule ≥2-3
 //     phases before its first ds_read (ledger in the loop comments).
 //   * fragment ds_reads run ONE phase ahead of their MFMA consumer

@@ -202,7 +202,7 @@ class Linear(Module):
         if need_dx:
             dx = F.linear_dgrad(dout, self.weight.compute(),
                                 self.weight.compute_t(), mask_src)
-        if self._defer_wgrad and dout.is_cuda:
+        if self._defer_wgrad:
             self._wgrad_pending.append((dout, x, mask_src))
             if len(self._wgrad_pending) >= self._wgrad_window:
                 self.flush_wgrad()
@@ -379,6 +379,14 @@ class Sequential(Module):
         # dL/d(data) — never consumed — so its dgrad GEMM is skipped
         # (the most expensive dgrad: K = input width).
         self._skip_input_grad = False
+        # set by the Worker for the temporally-LAST backward of a batch
+        # under deferred wgrad: each layer's pending µbatch wgrad chunks
+        # flush the moment that layer's backward completes, so its grads
+        # are final MID-backward and the DP bucket all-reduce fired from
+        # the grad hooks overlaps the remaining layers' dgrad/wgrad
+        # kernels (reference pipe.py:302-316 semantics, restored for the
+        # deferred path — round-1 deferred all grads to OptimizerStep).
+        self._flush_in_backward = False
 
     def forward(self, inputs, mubatch_id: int = 0):
         x = inputs
@@ -394,6 +402,11 @@ class Sequential(Module):
                 d = layer.backward(d, mubatch_id, need_dx=False)
             else:
                 d = layer.backward(d, mubatch_id)
+            # deferred-wgrad final backward: complete this layer's grads
+            # NOW (before its grad hooks fire) so param_done sees final
+            # values and the bucket all-reduce launches mid-backward
+            if self._flush_in_backward and hasattr(layer, "flush_wgrad"):
+                layer.flush_wgrad()
             for hook in self._grad_hooks:
                 for p in layer.parameters():
                     if p.requires_grad:
@@ -447,15 +460,20 @@ class Sequential(Module):
                 layer._wgrad_window = max(1, window)
 
     def flush_wgrads(self, per_layer_hook=None):
-        """Flush deferred weight gradients layer by layer in BACKWARD
-        order (so DP bucket all-reduces of later layers overlap the
-        remaining wgrad kernels); per_layer_hook(layer) fires after
-        each layer's grads are final."""
+        """Flush any REMAINING deferred weight gradients layer by layer
+        in backward order; returns how many layers still had pending
+        chunks.  On the normal path this is a no-op safety net — the
+        final (AllReduce) backward flushes in-backward via
+        _flush_in_backward; leftovers here mean a schedule never issued
+        BackwardGradAllReduce."""
+        n = 0
         for layer in reversed(self.layers):
-            if hasattr(layer, "flush_wgrad"):
+            if hasattr(layer, "flush_wgrad") and layer._wgrad_pending:
                 layer.flush_wgrad()
+                n += 1
             if per_layer_hook is not None:
                 per_layer_hook(layer)
+        return n
 
     def materialize_device(self, device, compute_dtype=None):
         """Move to device, set compute dtype, and re-point every

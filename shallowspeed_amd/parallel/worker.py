@@ -37,6 +37,16 @@ from .instructions import (
 )
 
 
+class _BatchedP2P:
+    """A run of >=2 consecutive p2p instructions coalesced into one
+    grouped RCCL call (see Worker._coalesce_p2p)."""
+
+    __slots__ = ("cmds",)
+
+    def __init__(self, cmds):
+        self.cmds = cmds
+
+
 class Worker:
     def __init__(self, topo: Topology, model, dataset=None, optimizer=None,
                  use_dp: bool = True, bucket_bytes=None):
@@ -115,7 +125,8 @@ class Worker:
         # deferred-µbatch wgrad: one chunked kernel per layer at the
         # optimizer step instead of a wgrad launch per µbatch
         self._defer_active = (
-            self.device.type == "cuda"
+            (self.device.type == "cuda"
+             or getattr(self, "_force_defer", False))  # CPU test hook
             and getattr(schedule, "is_training", True)
             and schedule.num_micro_batches > 1
             and hasattr(self.model, "set_defer_wgrad")
@@ -141,8 +152,14 @@ class Worker:
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
+        batch_ok = self.topo.is_distributed and comm_mod.can_batch_p2p()
         for commands in schedule.steps():
+            if batch_ok:
+                commands = self._coalesce_p2p(commands)
             for cmd in commands:
+                if isinstance(cmd, _BatchedP2P):
+                    self._run_batched_p2p(cmd)
+                    continue
                 fn = self._DISPATCH[type(cmd)]
                 if self._timing:
                     with StepTimer(self.device) as t:
@@ -154,6 +171,57 @@ class Worker:
                 else:
                     fn(cmd)
         self.flush_sends()
+
+    # ------------------------------------------- batched p2p (RCCL path)
+    # Runs of >=2 consecutive p2p instructions within one schedule step
+    # (e.g. send-activations-to-next + recv-grad-from-next — the same
+    # peer, both directions) become ONE grouped ncclSend/Recv call via
+    # dist.batch_isend_irecv: NCCL's matched-order group semantics make
+    # the bidirectional edge robust regardless of per-call enqueue
+    # interleaving across ranks.  Single p2p instructions keep the
+    # (cheaper) per-op path.
+    _P2P_TYPES = (SendActivations, SendInputGrad,
+                  RecvActivations, RecvOutputGrad)
+
+    def _coalesce_p2p(self, commands):
+        out, run = [], []
+        for cmd in commands:
+            if isinstance(cmd, self._P2P_TYPES):
+                run.append(cmd)
+                continue
+            if run:
+                out.append(_BatchedP2P(run) if len(run) > 1 else run[0])
+                run = []
+            out.append(cmd)
+        if run:
+            out.append(_BatchedP2P(run) if len(run) > 1 else run[0])
+        return out
+
+    def _run_batched_p2p(self, batch):
+        ops = []
+        for cmd in batch.cmds:
+            if isinstance(cmd, SendActivations):
+                self._wait_buffer("out", cmd.buffer_idx)
+                ops.append(("send", self._out_bufs[cmd.buffer_idx],
+                            self.topo.next_rank))
+            elif isinstance(cmd, SendInputGrad):
+                self._wait_buffer("gout", cmd.buffer_idx)
+                ops.append(("send", self._gout_bufs[cmd.buffer_idx],
+                            self.topo.prev_rank))
+            elif isinstance(cmd, RecvActivations):
+                self._in_views.pop(cmd.buffer_idx, None)
+                self._wait_buffer("in", cmd.buffer_idx)
+                ops.append(("recv", self._in_bufs[cmd.buffer_idx],
+                            self.topo.prev_rank))
+            else:  # RecvOutputGrad
+                self._wait_buffer("gin", cmd.buffer_idx)
+                ops.append(("recv", self._gin_bufs[cmd.buffer_idx],
+                            self.topo.next_rank))
+        for w in comm_mod.batch_p2p(ops):
+            # NCCL: stream-level wait (compute after this sees the
+            # recv'd data; the send buffers become reusable in stream
+            # order — no handle parking needed for batched sends)
+            w.wait()
 
     def enable_instruction_timing(self, on: bool = True):
         self._timing = on
@@ -252,25 +320,16 @@ class Worker:
 
     def _optimizer_step(self, cmd):
         if getattr(self, "_defer_active", False):
-            self._flush_deferred_wgrads()
+            # normally a no-op: the AllReduce backward flushed every
+            # layer mid-backward (_flush_in_backward).  Leftovers mean
+            # the schedule never issued BackwardGradAllReduce — under
+            # DP that would desync replicas, so fail loudly.
+            n = self.model.flush_wgrads()
+            if n and self.reducer is not None:
+                raise RuntimeError(
+                    f"{n} layers' deferred wgrads flushed AFTER the DP "
+                    f"reduction — schedule issued no BackwardGradAllReduce")
         self.optimizer.step()
-
-    def _flush_deferred_wgrads(self):
-        """Chunked wgrad flush in backward layer order; in DP mode the
-        bucket all-reduces fire here (grads only become final now) and
-        overlap the remaining layers' wgrad kernels."""
-        if self.reducer is not None:
-            self.reducer.reset()
-
-            def hook(layer):
-                for p in layer.parameters():
-                    if p.requires_grad:
-                        self.reducer.param_done(p)
-
-            self.model.flush_wgrads(hook)
-            self.reducer.finalize()
-        else:
-            self.model.flush_wgrads()
 
     def _load_input(self, cmd):
         if getattr(self, "_use_staged", False):
@@ -359,10 +418,15 @@ class Worker:
         'inversion' in the stack (pipe.py:389-400 ↔ layers.py:201-213):
         grad-ready → bucket all-reduce launches mid-backward.
 
-        In deferred-wgrad mode grads are not final until the flush at
-        OptimizerStep, so the hooks move there."""
+        Deferred-wgrad mode: every training schedule places
+        BackwardGradAllReduce on the temporally LAST backward of the
+        batch (naive: last µbatch; GPipe: µbatch 0, processed last;
+        1F1B: last cooldown backward), so _flush_in_backward makes each
+        layer's grads final the moment that layer's backward completes
+        — its bucket's all-reduce then overlaps the remaining layers'
+        dgrad/wgrad kernels, same as the eager path."""
         if getattr(self, "_defer_active", False):
-            return self._backward_acc(cmd)
+            self.model._flush_in_backward = True
         if self.reducer is not None:
             self.reducer.reset()
             self.model.register_grad_hook(self.reducer.param_done)
@@ -371,6 +435,7 @@ class Worker:
         try:
             self._backward_acc(cmd)
         finally:
+            self.model._flush_in_backward = False
             if self.reducer is not None:
                 self.model.reset_grad_hooks()
                 self.model.reset_post_grad_hooks()

@@ -57,7 +57,7 @@ def _run_dist(fn, world, tmp_path, **kwargs):
 # ------------------------------------------------------------------ DP
 
 def _dp_train(rank, world, out_dir, loss="mse", schedule="naive",
-              bucket_bytes=25 << 20):
+              bucket_bytes=25 << 20, force_defer=False):
     from shallowspeed_amd.parallel import init_topology
 
     topo = init_topology(dp=world, pp=1, backend="gloo",
@@ -71,9 +71,26 @@ def _dp_train(rank, world, out_dir, loss="mse", schedule="naive",
     w = Worker(topo, model, ds, opt, bucket_bytes=bucket_bytes)
     if bucket_bytes < (1 << 20):
         assert len(w.reducer.buckets) > 1, "test wants multiple buckets"
+    seen = {"in_bwd": 0, "elsewhere": 0}
+    if force_defer:
+        # exercise the deferred-wgrad path (normally GPU-only) and
+        # assert the bucket all-reduce hooks fire MID-backward (the
+        # restored reference pipe.py:302-316 overlap), never at the
+        # optimizer-step flush
+        w._force_defer = True
+        orig = w.reducer.param_done
+
+        def spy(p):
+            key = "in_bwd" if model._flush_in_backward else "elsewhere"
+            seen[key] += 1
+            return orig(p)
+
+        w.reducer.param_done = spy
     cls = SCHEDULES[schedule]
     for b in range(ds.num_batches()):
         w.execute(cls(ds.num_mubatches(), 1, 0), b)
+    if force_defer:
+        assert seen["in_bwd"] > 0 and seen["elsewhere"] == 0, seen
     # replicas must hash-identical (reference train.py:154-155)
     assert_sync(topo.dp_group, get_model_hash(model))
     if rank == 0:
@@ -96,6 +113,19 @@ def test_dp2_multibucket_matches_serial(tmp_path):
     """Small bucket size => several overlapped all-reduces per backward
     (the bucketed path with >1 bucket)."""
     _run_dist(_dp_train, 2, tmp_path, schedule="gpipe", bucket_bytes=2048)
+    got = torch.load(tmp_path / "dp_params.pt", weights_only=False)
+    want = _serial_params()
+    for g, w in zip(got, want):
+        torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.parametrize("schedule", ["naive", "gpipe", "pipedream"])
+def test_dp2_deferred_wgrad_matches_serial(tmp_path, schedule):
+    """Deferred-µbatch wgrad (the GPU hot path, forced on CPU): grads
+    flush mid-final-backward, bucket all-reduces overlap the remaining
+    backward, replicas stay in sync and match serial training."""
+    _run_dist(_dp_train, 2, tmp_path, schedule=schedule,
+              bucket_bytes=2048, force_defer=True)
     got = torch.load(tmp_path / "dp_params.pt", weights_only=False)
     want = _serial_params()
     for g, w in zip(got, want):

@@ -41,11 +41,18 @@ def main():
                          "faster at the flagship step size — measured)")
     ap.add_argument("--no-graph", action="store_true",
                     help="(compat) force eager; eager is the default")
+    ap.add_argument("--deterministic", action="store_true",
+                    help="bitwise-reproducible mode (split_k=1 wgrad)")
     args = ap.parse_args()
 
     from shallowspeed_amd.data import Dataset
     from shallowspeed_amd.models import MLP, SGD
     from shallowspeed_amd.parallel import SCHEDULES, Worker, init_topology
+
+    if args.deterministic:
+        from shallowspeed_amd.ops import functional as _F
+
+        _F.set_deterministic(True)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     assert world == args.gpus or args.gpus == 1 or world == 1, \
@@ -99,7 +106,10 @@ def main():
         if is_gpu:
             torch.cuda.synchronize(device)
         if topo.is_distributed:
-            dist.barrier()
+            if is_gpu and dist.get_backend() == "nccl":
+                dist.barrier(device_ids=[device.index])
+            else:
+                dist.barrier()
         if is_gpu:
             torch.cuda.synchronize(device)
 

@@ -8,7 +8,15 @@ DP rank 0 of that stage (every DP replica is identical — enforced by
 assert_sync).  Layout:
 
     <dir>/meta.pt                      (world/dp/pp, sizes, step)
-    <dir>/stage_{s:02d}.pt             (f32 master params, in order)
+    <dir>/stage_{s:02d}.pt             (f32 master params + optimizer
+                                        state — momentum velocities —
+                                        in parameter order)
+
+Resuming a momentum>0 run restores the exact optimization trajectory
+(velocities + hyperparameters are saved with the stage shard); loading
+validates the checkpoint's PP topology against the current one so a
+repartitioned resume fails loudly instead of via per-tensor shape
+asserts alone.
 """
 
 import os
@@ -16,7 +24,8 @@ import os
 import torch
 
 
-def save_checkpoint(path, model, topo, step: int = 0, extra=None):
+def save_checkpoint(path, model, topo, step: int = 0, extra=None,
+                    optimizer=None):
     os.makedirs(path, exist_ok=True)
     if topo.dp_rank == 0:
         state = {
@@ -24,6 +33,13 @@ def save_checkpoint(path, model, topo, step: int = 0, extra=None):
             "stage_id": topo.stage_id,
             "step": step,
         }
+        if optimizer is not None and getattr(optimizer, "_vel", None) is not None:
+            state["opt"] = {
+                "velocity": [v.detach().cpu() for v in optimizer._vel],
+                "lr": optimizer.lr,
+                "momentum": optimizer.momentum,
+                "weight_decay": optimizer.weight_decay,
+            }
         torch.save(state, os.path.join(path, f"stage_{topo.stage_id:02d}.pt"))
     if topo.rank == 0:
         meta = {"dp": topo.dp, "pp": topo.pp, "step": step}
@@ -32,7 +48,15 @@ def save_checkpoint(path, model, topo, step: int = 0, extra=None):
         torch.save(meta, os.path.join(path, "meta.pt"))
 
 
-def load_checkpoint(path, model, topo):
+def load_checkpoint(path, model, topo, optimizer=None):
+    meta_f = os.path.join(path, "meta.pt")
+    meta = torch.load(meta_f, map_location="cpu", weights_only=False) \
+        if os.path.exists(meta_f) else {}
+    if "pp" in meta:
+        assert meta["pp"] == topo.pp, (
+            f"checkpoint was written with pp={meta['pp']}; current "
+            f"topology has pp={topo.pp} — stage shards do not repartition"
+        )
     f = os.path.join(path, f"stage_{topo.stage_id:02d}.pt")
     state = torch.load(f, map_location="cpu", weights_only=False)
     params = model.parameters()
@@ -44,7 +68,17 @@ def load_checkpoint(path, model, topo):
         assert p.data.shape == saved.shape, (p.data.shape, saved.shape)
         p.data.copy_(saved.to(p.data.device))
         p.sync_lp()
-    meta_f = os.path.join(path, "meta.pt")
-    meta = torch.load(meta_f, map_location="cpu", weights_only=False) \
-        if os.path.exists(meta_f) else {}
+    if optimizer is not None and "opt" in state:
+        opt_state = state["opt"]
+        vel = opt_state["velocity"]
+        if optimizer._vel is None:
+            # optimizer constructed with momentum=0 but checkpoint has
+            # velocities: adopt the saved hyperparameters
+            optimizer.momentum = opt_state["momentum"]
+            optimizer._vel = [torch.zeros_like(p.data)
+                              for p in optimizer.params]
+        assert len(vel) == len(optimizer._vel), (len(vel), len(optimizer._vel))
+        for dst, src in zip(optimizer._vel, vel):
+            assert dst.shape == src.shape, (dst.shape, src.shape)
+            dst.copy_(src.to(dst.device))
     return meta

@@ -483,6 +483,7 @@ class Sequential(Module):
         device = torch.device(device)
         if compute_dtype is None:
             compute_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+        F.clear_wgrad_tables()  # grad buffers are about to be repointed
         self.device = device
         self.compute_dtype = compute_dtype
         for layer in self.layers:

@@ -23,6 +23,29 @@ import torch
 from ._ext import load_ext
 
 
+_DETERMINISTIC = False
+
+
+def set_deterministic(flag: bool):
+    """Bitwise run-to-run reproducible GPU training.
+
+    Forces the single-owner (split_k=1) wgrad path so f32 atomicAdd
+    ARRIVAL ORDER cannot perturb gradient bits; all other hot kernels
+    (fwd/dgrad GEMMs, fused SGD, loss heads) are deterministic by
+    construction (fixed tile ownership, no atomics), and the DP bucket
+    all-reduce order is fixed by construction in GradReducer.  Matches
+    the reference's exact-equality replica gate (train.py:154-155) at
+    the bitwise level on GPU.  Costs wgrad parallelism on small-M
+    shapes (split-K is what fills the chip there) — opt-in via
+    --deterministic."""
+    global _DETERMINISTIC
+    _DETERMINISTIC = bool(flag)
+
+
+def deterministic() -> bool:
+    return _DETERMINISTIC
+
+
 def _is_gpu(t: torch.Tensor) -> bool:
     return t.is_cuda
 
@@ -105,9 +128,11 @@ def linear_wgrad_acc(dy, x, grad_w, grad_b=None, mask_src=None, split_k=0):
     split-K parallelism work: the GPU kernel uses f32 atomicAdd, so a
     K-split over the batch dimension and accumulation across µbatches
     compose for free.  split_k=0 lets the kernel pick; split_k=1 is the
-    deterministic single-owner path.
+    deterministic single-owner path (forced by set_deterministic).
     Reference: functional.py:21 (dW = dout.T @ x, db = dout.sum(0)).
     """
+    if split_k == 0 and _DETERMINISTIC:
+        split_k = 1
     if _is_gpu(dy):
         ext = _ext_for(dy)
         ext.wgrad_tn(
@@ -255,7 +280,21 @@ def gelu_bwd(dy, z):
     return zf.grad.to(dy.dtype)
 
 
-_WGRAD_TABLES = {}
+from collections import OrderedDict  # noqa: E402
+
+# LRU-capped cache of (pinned-host, device) pointer tables for the
+# chunked wgrad kernel, keyed by (grad_w ptr, n_chunks).  The cap plus
+# clear_wgrad_tables() (called from Sequential.materialize_device)
+# keeps long-lived processes that build/destroy many models from
+# leaking pinned+device entries; a freed-then-reallocated grad buffer
+# at the same address is safe regardless (rows are re-checked), the
+# eviction is purely about memory growth.
+_WGRAD_TABLES = OrderedDict()
+_WGRAD_TABLES_CAP = 64
+
+
+def clear_wgrad_tables():
+    _WGRAD_TABLES.clear()
 
 
 def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
@@ -267,6 +306,8 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
     schedules (each launch re-pays prologue + atomic epilogue).
     GPU only; the CPU path loops (it has no launch cost to amortize).
     """
+    if split_k == 0 and _DETERMINISTIC:
+        split_k = 1
     dy0 = chunks[0][0]
     if not _is_gpu(dy0):
         for dy, x, m in chunks:
@@ -288,11 +329,15 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
     key = (grad_w.data_ptr(), len(chunks))
     entry = _WGRAD_TABLES.get(key)
     if entry is None:
+        while len(_WGRAD_TABLES) >= _WGRAD_TABLES_CAP:
+            _WGRAD_TABLES.popitem(last=False)
         pinned = torch.empty(len(chunks), 3, dtype=torch.int64,
                              pin_memory=True)
         dev = torch.empty(len(chunks), 3, dtype=torch.int64,
                           device=dy0.device)
         entry = _WGRAD_TABLES[key] = [pinned, dev, None]
+    else:
+        _WGRAD_TABLES.move_to_end(key)
     pinned, table, last_rows = entry
     if rows != last_rows:
         # chunk pointers are usually STABLE step to step (caching

@@ -71,3 +71,49 @@ def test_resume_continues_identically(tmp_path):
 
     for a, b in zip(straight.parameters(), part2.parameters()):
         torch.testing.assert_close(a.data, b.data, rtol=0, atol=0)
+
+
+def _train_momentum(model, opt, ds, batches):
+    w = Worker(Topology(), model, ds, opt)
+    for b in batches:
+        w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
+
+
+def test_resume_with_momentum_restores_velocity(tmp_path):
+    """Momentum velocities are part of the checkpoint: resuming a
+    momentum>0 run continues the exact optimization trajectory
+    (round-1 gap: velocities were silently reset on resume)."""
+    mk_ds = lambda: Dataset(16, 8, n_samples=64, in_dim=SIZES[0],
+                            n_classes=SIZES[-1]).load(0, 1)
+
+    straight = MLP(SIZES, 0, 1, 16).materialize_device("cpu")
+    opt_s = SGD(straight.parameters(), lr=0.05, momentum=0.9)
+    _train_momentum(straight, opt_s, mk_ds(), [0, 1, 2, 3])
+
+    part1 = MLP(SIZES, 0, 1, 16).materialize_device("cpu")
+    opt1 = SGD(part1.parameters(), lr=0.05, momentum=0.9)
+    ds = mk_ds()
+    _train_momentum(part1, opt1, ds, [0, 1])
+    assert any(v.abs().sum() > 0 for v in opt1._vel)
+    save_checkpoint(tmp_path, part1, Topology(), step=2, optimizer=opt1)
+
+    part2 = MLP(SIZES, 0, 1, 16).materialize_device("cpu")
+    opt2 = SGD(part2.parameters(), lr=0.05, momentum=0.9)
+    load_checkpoint(tmp_path, part2, Topology(), optimizer=opt2)
+    for a, b in zip(opt1._vel, opt2._vel):
+        torch.testing.assert_close(a, b, rtol=0, atol=0)
+    _train_momentum(part2, opt2, ds, [2, 3])
+
+    for a, b in zip(straight.parameters(), part2.parameters()):
+        torch.testing.assert_close(a.data, b.data, rtol=0, atol=0)
+
+
+def test_load_rejects_mismatched_pp(tmp_path):
+    model = MLP(SIZES, 0, 1, 16).materialize_device("cpu")
+    save_checkpoint(tmp_path, model, Topology(), step=1)
+    import pytest
+
+    bad_topo = Topology(rank=0, world=2, dp=1, pp=2)
+    stage0 = MLP([SIZES[0], SIZES[1]], 0, 1, 16).materialize_device("cpu")
+    with pytest.raises(AssertionError, match="pp="):
+        load_checkpoint(tmp_path, stage0, bad_topo)

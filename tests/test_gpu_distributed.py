@@ -162,6 +162,47 @@ def test_pp2_on_gpu_matches_serial(tmp_path, gpu_device, schedule):
         torch.testing.assert_close(g, w, atol=5e-3, rtol=5e-2)
 
 
+def _dp_gpu_det(rank, world, out_dir, tag="a"):
+    """DP training with --deterministic semantics: split_k=1 wgrad +
+    fixed bucket order.  Saves the final model hash for bitwise
+    comparison across runs."""
+    from shallowspeed_amd.data import Dataset
+    from shallowspeed_amd.models import MLP, SGD
+    from shallowspeed_amd.ops import functional as F
+    from shallowspeed_amd.parallel import SCHEDULES, Worker, init_topology
+    from shallowspeed_amd.utils import assert_sync, get_model_hash
+
+    F.set_deterministic(True)
+    dev = torch.device("cuda", 0)
+    topo = init_topology(dp=world, pp=1, backend="gloo", device=dev)
+    model = MLP(SIZES, 0, 1, GBS, loss="mse").materialize_device(dev)
+    opt = SGD(model.parameters(), lr=LR)
+    ds = Dataset(GBS, (GBS // world) // MUB, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1], device=dev).load(topo.dp_rank, world)
+    w = Worker(topo, model, ds, opt)
+    for b in range(ds.num_batches()):
+        w.execute(SCHEDULES["gpipe"](ds.num_mubatches(), 1, 0), b)
+    torch.cuda.synchronize()
+    # bitwise replica equality (the reference's exact-equality gate,
+    # train.py:154-155, at full strength on GPU)
+    assert_sync(topo.dp_group, get_model_hash(model))
+    if rank == 0:
+        with open(os.path.join(out_dir, f"det_hash_{tag}"), "w") as f:
+            f.write(get_model_hash(model))
+    torch.distributed.destroy_process_group()
+
+
+def test_deterministic_dp2_bitwise_reproducible(tmp_path, gpu_device):
+    """--deterministic mode: two identical DP=2 trainings produce
+    BITWISE-identical models (sha1 hash equality), and replicas are
+    bitwise in sync — no 5e-3 tolerances."""
+    _run(_dp_gpu_det, 2, tmp_path, tag="a")
+    _run(_dp_gpu_det, 2, tmp_path, tag="b")
+    ha = (tmp_path / "det_hash_a").read_text()
+    hb = (tmp_path / "det_hash_b").read_text()
+    assert ha == hb, (ha, hb)
+
+
 def _pp_mem(rank, world, out_dir, schedule="gpipe"):
     from shallowspeed_amd.data import Dataset
     from shallowspeed_amd.models import MLP, SGD

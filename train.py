@@ -70,6 +70,12 @@ def main():
     ap.add_argument("--global-batch", type=int, default=128)
     ap.add_argument("--mubatches", type=int, default=4)
     ap.add_argument("--lr", type=float, default=0.006)
+    ap.add_argument("--momentum", type=float, default=0.0)
+    ap.add_argument("--weight-decay", type=float, default=0.0)
+    ap.add_argument("--deterministic", action="store_true",
+                    help="bitwise run-to-run reproducible GPU training "
+                         "(forces the single-owner split_k=1 wgrad path; "
+                         "bucket reduction order is fixed by construction)")
     ap.add_argument("--layer-sizes", type=parse_sizes,
                     default=[784, 128, 127, 126, 125, 124, 123, 10],
                     help="comma-separated boundaries (reference train.py:98)")
@@ -99,15 +105,21 @@ def main():
     device = topo.device
 
     # model: stage slice of the full MLP (reference train.py:99-107)
+    if args.deterministic:
+        from shallowspeed_amd.ops import functional as _F
+
+        _F.set_deterministic(True)
+
     model = MLP(args.layer_sizes, stage_idx=topo.stage_id, n_stages=args.pp,
                 global_batch_size=args.global_batch, loss=args.loss)
     model.materialize_device(device)
-    optimizer = SGD(model.parameters(), lr=args.lr)
+    optimizer = SGD(model.parameters(), lr=args.lr, momentum=args.momentum,
+                    weight_decay=args.weight_decay)
 
     if args.resume:
         from shallowspeed_amd.checkpoint import load_checkpoint
 
-        load_checkpoint(args.resume, model, topo)
+        load_checkpoint(args.resume, model, topo, optimizer=optimizer)
 
     mubatch = args.global_batch // args.dp // args.mubatches
     train_ds = Dataset(args.global_batch, mubatch, save_dir=args.data_dir,
@@ -150,7 +162,8 @@ def main():
     if args.save:
         from shallowspeed_amd.checkpoint import save_checkpoint
 
-        save_checkpoint(args.save, model, topo, step=args.epochs)
+        save_checkpoint(args.save, model, topo, step=args.epochs,
+                        optimizer=optimizer)
 
     if args.timing and topo.rank == 0:
         total = sum(worker.instruction_times.values())

@@ -85,6 +85,9 @@ def main():
                     help="load x_*.parquet / y_*.npy like the reference "
                          "if present; otherwise synthetic")
     ap.add_argument("--samples", type=int, default=8192)
+    ap.add_argument("--val-batch", type=int, default=None,
+                    help="validation batch size (default: auto — whole "
+                         "val split when --data-dir files exist)")
     ap.add_argument("--backend", default=None,
                     help="torch.distributed backend override "
                          "(default: nccl on GPU, gloo on CPU)")
@@ -126,7 +129,18 @@ def main():
                        n_samples=args.samples, in_dim=args.layer_sizes[0],
                        n_classes=args.layer_sizes[-1], device=device)
     train_ds.load(topo.dp_rank, args.dp)
-    val_batch = min(args.samples // 4, 1024)
+    val_batch = args.val_batch
+    if val_batch is None and args.data_dir:
+        import os
+
+        yp = os.path.join(args.data_dir, "y_val.npy")
+        if os.path.exists(yp):
+            import numpy as np
+
+            n_val = np.load(yp, mmap_mode="r").shape[0]
+            val_batch = max(1, min(n_val // args.dp, 2048))
+    if val_batch is None:
+        val_batch = min(args.samples // 4, 1024)
     val_ds = Dataset(val_batch * args.dp, val_batch, save_dir=args.data_dir,
                      validation=True, n_samples=max(args.samples // 4, val_batch * args.dp),
                      in_dim=args.layer_sizes[0],

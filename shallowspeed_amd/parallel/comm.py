@@ -247,10 +247,17 @@ def recv_tensor(t: torch.Tensor, src_rank: int):
     dist.recv(t, src=src_rank)
 
 
-def can_batch_p2p() -> bool:
-    """Batched p2p is used on the RCCL device path only (the gloo/CPU
-    test path keeps the per-op staging fallback)."""
-    return dist.is_initialized() and dist.get_backend() == "nccl"
+def can_batch_p2p(device=None) -> bool:
+    """Batched p2p runs on the RCCL device path and on gloo with CPU
+    tensors (so the multi-process CPU tests exercise the same batched
+    code path the GPU uses).  gloo with CUDA tensors keeps the per-op
+    CPU-staging fallback."""
+    if not dist.is_initialized():
+        return False
+    backend = dist.get_backend()
+    if backend == "nccl":
+        return True
+    return backend == "gloo" and device is not None and device.type == "cpu"
 
 
 def batch_p2p(ops):

@@ -152,7 +152,8 @@ class Worker:
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
-        batch_ok = self.topo.is_distributed and comm_mod.can_batch_p2p()
+        batch_ok = self.topo.is_distributed and \
+            comm_mod.can_batch_p2p(self.device)
         for commands in schedule.steps():
             if batch_ok:
                 commands = self._coalesce_p2p(commands)

@@ -189,10 +189,22 @@ class Linear(Module):
                 # WIDE aligned layers: materializing dz once lets
                 # dgrad take the unmasked glds tier and strips the
                 # mask re-reads from wgrad (mask fusion wins at small
-                # shapes where L2 absorbs the re-reads — measured)
-                if (B % 128 == 0 and I % 128 == 0 and O % 32 == 0
-                        and (B // 128) * (I // 128) >= 512
-                        and O >= 512 and I >= 512):
+                # shapes where L2 absorbs the re-reads — measured).
+                # Second disjunct: any shape the 256²-tile wgrad tier
+                # can take (O%256, I%256, B%128) unmasks too — the
+                # masked path would silently fall back to the 128²
+                # wgrad tier (~0.65 PF vs 1.1-1.2 PF), which costs far
+                # more than one elementwise dz pass.  dz is shared by
+                # dgrad and (deferred) wgrad, so the pass amortizes
+                # over both GEMMs.
+                wide_glds = (B % 128 == 0 and I % 128 == 0 and O % 32 == 0
+                             and (B // 128) * (I // 128) >= 512
+                             and O >= 512 and I >= 512)
+                wgrad256_ok = (O % 256 == 0 and I % 256 == 0
+                               and B % 128 == 0 and B >= 128
+                               and O >= 512 and I >= 512
+                               and (O // 256) * (I // 256) >= 64)
+                if wide_glds or wgrad256_ok:
                     dout = F.relu_bwd(dout, mask_src)
                     mask_src = None
         elif self.activation == "gelu":

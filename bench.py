@@ -43,6 +43,8 @@ def main():
                     help="(compat) force eager; eager is the default")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise-reproducible mode (split_k=1 wgrad)")
+    ap.add_argument("--fp8-fwd", action="store_true",
+                    help="MX-fp8 training forward on qualifying layers")
     args = ap.parse_args()
 
     from shallowspeed_amd.data import Dataset
@@ -79,6 +81,8 @@ def main():
 
     model = MLP(sizes, topo.stage_id, pp, global_batch, loss=args.loss)
     model.materialize_device(device)
+    if args.fp8_fwd:
+        model.set_fp8_fwd(True)
     opt = SGD(model.parameters(), lr=0.001)
     # synthetic data of the benchmark shape, random-init weights (no
     # network for datasets/checkpoints in this environment)
@@ -157,6 +161,7 @@ def main():
                 "loss": args.loss,
                 "mubatches": args.mubatches,
                 "hipgraph": use_graph,
+                "fp8_fwd": args.fp8_fwd,
             },
         }), flush=True)
 

@@ -159,9 +159,33 @@ class Linear(Module):
         self._defer_wgrad = False
         self._wgrad_window = 1 << 30
         self._wgrad_pending = []
+        # MX-fp8 forward (training): quantize activations per µbatch
+        # and keep an fp8 copy of the weight, refreshed lazily after
+        # each optimizer step (invalidate_fp8()).  Backward stays bf16
+        # (dgrad/wgrad on the bf16 stash); the f32 master is untouched
+        # — this is the training-forward extension of the round-1
+        # serving tier (csrc/fp8.hip, ~1.7x the bf16 8-phase GEMM).
+        self.fp8_fwd = False
+        self._wq = None
+        self._ws = None
+
+    def _fp8_ok(self, rows: int) -> bool:
+        return (self.fp8_fwd and self.activation in (None, "relu")
+                and rows % 256 == 0
+                and self.out_dims % 256 == 0 and self.in_dims % 256 == 0)
 
     def forward(self, inputs, mubatch_id: int = 0):
         self._stash("x", mubatch_id, inputs)
+        if inputs.is_cuda and self._fp8_ok(inputs.shape[0]):
+            if self._wq is None:
+                self._wq, self._ws = F.fp8_quantize(self.weight.compute())
+            xq, xs = F.fp8_quantize(inputs)
+            y = F.linear_fwd_fp8(xq, xs, self._wq, self._ws,
+                                 self.bias.compute(),
+                                 relu=(self.activation == "relu"))
+            if self.activation == "relu":
+                self._stash("y", mubatch_id, y)
+            return y
         if self.activation == "gelu":
             z = F.linear_fwd(inputs, self.weight.compute(),
                              self.bias.compute(), relu=False)
@@ -459,6 +483,26 @@ class Sequential(Module):
         self._training = False
         for l in self.layers:
             l.eval()
+
+    def set_fp8_fwd(self, flag: bool):
+        """Enable the MX-fp8 training forward on every qualifying
+        Linear (O%256, I%256; per-call µbatch%256 check in _fp8_ok).
+        Backward stays bf16; masters stay f32."""
+        n = 0
+        for layer in self.layers:
+            if isinstance(layer, Linear) and layer.activation in (None, "relu") \
+                    and layer.out_dims % 256 == 0 and layer.in_dims % 256 == 0:
+                layer.fp8_fwd = flag
+                layer._wq = layer._ws = None
+                n += 1
+        return n
+
+    def invalidate_fp8(self):
+        """Drop cached fp8 weight copies (call after every optimizer
+        step / checkpoint load so the next forward re-quantizes)."""
+        for layer in self.layers:
+            if getattr(layer, "_wq", None) is not None:
+                layer._wq = layer._ws = None
 
     def set_defer_wgrad(self, flag: bool, window: int = 1 << 30):
         """Enable/disable deferred-µbatch wgrad on every Linear

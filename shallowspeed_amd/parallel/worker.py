@@ -331,6 +331,8 @@ class Worker:
                     f"{n} layers' deferred wgrads flushed AFTER the DP "
                     f"reduction — schedule issued no BackwardGradAllReduce")
         self.optimizer.step()
+        if hasattr(self.model, "invalidate_fp8"):
+            self.model.invalidate_fp8()
 
     def _load_input(self, cmd):
         if getattr(self, "_use_staged", False):

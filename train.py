@@ -76,6 +76,10 @@ def main():
                     help="bitwise run-to-run reproducible GPU training "
                          "(forces the single-owner split_k=1 wgrad path; "
                          "bucket reduction order is fixed by construction)")
+    ap.add_argument("--fp8-fwd", action="store_true",
+                    help="MX-fp8 training forward on qualifying Linears "
+                         "(O%%256,I%%256,µbatch%%256); backward stays bf16, "
+                         "masters f32")
     ap.add_argument("--layer-sizes", type=parse_sizes,
                     default=[784, 128, 127, 126, 125, 124, 123, 10],
                     help="comma-separated boundaries (reference train.py:98)")
@@ -123,6 +127,10 @@ def main():
         from shallowspeed_amd.checkpoint import load_checkpoint
 
         load_checkpoint(args.resume, model, topo, optimizer=optimizer)
+
+    if args.fp8_fwd:
+        n = model.set_fp8_fwd(True)
+        rprint(f"MX-fp8 training forward enabled on {n} layer(s)")
 
     mubatch = args.global_batch // args.dp // args.mubatches
     train_ds = Dataset(args.global_batch, mubatch, save_dir=args.data_dir,

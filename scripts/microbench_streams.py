@@ -13,11 +13,15 @@ streams.
 Run (GPU): python scripts/microbench_streams.py
 """
 
+import os
+import sys
 import time
 
 import torch
 
-from shallowspeed_amd.ops import functional as F
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from shallowspeed_amd.ops import functional as F  # noqa: E402
 
 
 def fwd_chain(x, ws, bs):

@@ -17,6 +17,8 @@ RCCL/xGMI; the DP all-reduce rides torch.distributed's comm stream and
 overlaps backward kernels.
 """
 
+import contextlib
+
 import torch
 
 from ..utils import StepTimer
@@ -132,15 +134,6 @@ class Worker:
             and hasattr(self.model, "set_defer_wgrad")
             and not getattr(self, "_force_no_defer", False)
         )
-        if hasattr(self.model, "set_defer_wgrad"):
-            # window = schedule's in-flight bound, floored at 4:
-            # GPipe (O(M) stash by design) batches all µbatches; 1F1B
-            # keeps its ~warmup+1 memory bound to within a small
-            # constant (≤4 retained µbatch grad/act pairs) while still
-            # amortizing kernel launches 4x
-            win = max(4, getattr(schedule, "max_in_flight",
-                                 schedule.num_micro_batches))
-            self.model.set_defer_wgrad(self._defer_active, window=win)
         # snapshot recv'd/loaded activations when layers may retain
         # them past the buffer's reuse: >1 µbatch in flight, or wgrad
         # deferral holding them until flush.  (Zero-copy dataset VIEWS
@@ -149,9 +142,44 @@ class Worker:
             getattr(schedule, "max_in_flight", 2) > 1
             or self._defer_active
         )
+        # µbatch-parallel HIP streams (single-stage schedules only):
+        # on one stage, different µbatches' forwards/backwards are
+        # data-independent (grads accumulate via f32 atomics), but each
+        # small per-µbatch kernel pays a fixed wave-ramp/launch floor —
+        # the µbatch tax.  Round-robining µbatches over 2 side streams
+        # overlaps those ramps (the MI355X-idiomatic fix: fill the
+        # 256-CU chip from concurrent streams instead of serializing
+        # small kernels).  Gated off for pp>1 (stage p2p orders the
+        # pipeline) and for the AllReduce backward (joins first).
+        self._mu_par = (
+            self.device.type == "cuda"
+            and self.topo.pp == 1
+            and getattr(schedule, "is_training", True)
+            and schedule.num_micro_batches > 1
+            and not getattr(self, "_force_no_mu_par", False)
+        )
+        if self._mu_par and not hasattr(self, "_mu_streams"):
+            self._mu_streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+        if hasattr(self.model, "set_defer_wgrad"):
+            # window = schedule's in-flight bound, floored at 4:
+            # GPipe (O(M) stash by design) batches all µbatches; 1F1B
+            # keeps its ~warmup+1 memory bound to within a small
+            # constant (≤4 retained µbatch grad/act pairs) while still
+            # amortizing kernel launches 4x.  Under µbatch-parallel
+            # streams the window is the whole batch: an early window
+            # flush would launch on one side stream while reading
+            # chunks produced on the other (cross-stream race); with
+            # pp==1 there is no pipeline memory bound to protect.
+            if self._mu_par:
+                win = schedule.num_micro_batches
+            else:
+                win = max(4, getattr(schedule, "max_in_flight",
+                                     schedule.num_micro_batches))
+            self.model.set_defer_wgrad(self._defer_active, window=win)
         mub = self.dataset.mubatch_size if self.dataset is not None else \
             self._buf_shape[1]
         self._ensure_buffers(schedule.num_buffers, mub)
+        self._mu_fork()
         batch_ok = self.topo.is_distributed and \
             comm_mod.can_batch_p2p(self.device)
         for commands in schedule.steps():
@@ -315,11 +343,38 @@ class Worker:
             h.wait()
         self._pending_send.clear()
 
+    # ------------------------------------------- µbatch stream plumbing
+    # Strict fork/join phase discipline: side streams always WAIT the
+    # main stream at fork, main WAITS the side streams at join (before
+    # the AllReduce backward and the optimizer step).  Every
+    # cross-stream tensor handoff (stashes, deferred wgrad chunks read
+    # by the main-stream flush) crosses a join, and allocator reuse of
+    # side-allocated blocks happens only after the next fork — so the
+    # stream-aware caching allocator needs no record_stream calls.
+    def _mu_ctx(self, mubatch_id):
+        if not getattr(self, "_mu_par", False):
+            return contextlib.nullcontext()
+        return torch.cuda.stream(self._mu_streams[mubatch_id % 2])
+
+    def _mu_fork(self):
+        if getattr(self, "_mu_par", False):
+            cur = torch.cuda.current_stream()
+            for s in self._mu_streams:
+                s.wait_stream(cur)
+
+    def _mu_join(self):
+        if getattr(self, "_mu_par", False):
+            cur = torch.cuda.current_stream()
+            for s in self._mu_streams:
+                cur.wait_stream(s)
+
     # ------------------------------------------------- instruction impls
     def _zero_grad(self, cmd):
         self.model.zero_grad()
+        self._mu_fork()
 
     def _optimizer_step(self, cmd):
+        self._mu_join()
         if getattr(self, "_defer_active", False):
             # normally a no-op: the AllReduce backward flushed every
             # layer mid-backward (_flush_in_backward).  Leftovers mean
@@ -350,7 +405,8 @@ class Worker:
             return
         self._wait_buffer("in", cmd.buffer_idx)
         self._in_views.pop(cmd.buffer_idx, None)
-        buf.copy_(x.to(self.compute_dtype), non_blocking=True)
+        with self._mu_ctx(cmd.mubatch_id):
+            buf.copy_(x.to(self.compute_dtype), non_blocking=True)
 
     def _load_target(self, cmd):
         if getattr(self, "_use_staged", False):
@@ -361,7 +417,8 @@ class Worker:
         self._wait_buffer("gin", cmd.buffer_idx)
         buf = self._gin_bufs[cmd.buffer_idx]
         assert y.shape == buf.shape, (y.shape, buf.shape)  # pipe.py:362-365
-        buf.copy_(y.to(self.compute_dtype), non_blocking=True)
+        with self._mu_ctx(cmd.mubatch_id):
+            buf.copy_(y.to(self.compute_dtype), non_blocking=True)
 
     def _recv_activations(self, cmd):
         self._in_views.pop(cmd.buffer_idx, None)
@@ -385,36 +442,39 @@ class Worker:
         from_view = x is not None
         if x is None:
             x = self._in_bufs[cmd.in_buffer]
-        if self.model._training and self._snapshot_inputs and not from_view:
-            # Layers stash their input per µbatch for wgrad; the input
-            # buffer is SHARED across µbatches (overwritten by the next
-            # Load/Recv), so snapshot it when more than one µbatch can
-            # be in flight.  (The reference stashes the live buffer
-            # reference — layers.py:117 + pipe.py:447-454 — which is
-            # only safe for its naive schedule ordering.)
-            x = x.clone()
-        y = self.model.forward(x, cmd.mubatch_id)
-        if getattr(self, "_skip_out_copy", False):
-            return
-        self._wait_buffer("out", cmd.out_buffer)
-        self._out_bufs[cmd.out_buffer].copy_(y)
+        with self._mu_ctx(cmd.mubatch_id):
+            if self.model._training and self._snapshot_inputs and not from_view:
+                # Layers stash their input per µbatch for wgrad; the
+                # input buffer is SHARED across µbatches (overwritten by
+                # the next Load/Recv), so snapshot it when more than one
+                # µbatch can be in flight.  (The reference stashes the
+                # live buffer reference — layers.py:117 + pipe.py:447-454
+                # — which is only safe for its naive schedule ordering.)
+                x = x.clone()
+            y = self.model.forward(x, cmd.mubatch_id)
+            if getattr(self, "_skip_out_copy", False):
+                return
+            self._wait_buffer("out", cmd.out_buffer)
+            self._out_bufs[cmd.out_buffer].copy_(y)
 
-    def _backward_acc(self, cmd):
-        g = self._gin_bufs[cmd.out_buffer]
-        if getattr(self, "_defer_active", False) and \
-                self.topo.stage_id != self.topo.pp - 1:
-            # non-last stages hand the RECV'D grad buffer to the
-            # stage's last Linear, whose deferred wgrad would retain
-            # the reference past the buffer's next overwrite — snapshot
-            # it.  (The last stage's gin holds the TARGET, which the
-            # loss head consumes immediately.)
-            g = g.clone()
-        d = self.model.backward(g, cmd.mubatch_id)
-        # stage 0 never sends input grads, so skip the staging copy
-        if d is not None and self.topo.stage_id != 0:
-            self._wait_buffer("gout", cmd.in_buffer)
-            self._gout_bufs[cmd.in_buffer].copy_(d)
-        return d
+    def _backward_acc(self, cmd, _mu_stream_ok=True):
+        with (self._mu_ctx(cmd.mubatch_id) if _mu_stream_ok
+              else contextlib.nullcontext()):
+            g = self._gin_bufs[cmd.out_buffer]
+            if getattr(self, "_defer_active", False) and \
+                    self.topo.stage_id != self.topo.pp - 1:
+                # non-last stages hand the RECV'D grad buffer to the
+                # stage's last Linear, whose deferred wgrad would retain
+                # the reference past the buffer's next overwrite —
+                # snapshot it.  (The last stage's gin holds the TARGET,
+                # which the loss head consumes immediately.)
+                g = g.clone()
+            d = self.model.backward(g, cmd.mubatch_id)
+            # stage 0 never sends input grads, so skip the staging copy
+            if d is not None and self.topo.stage_id != 0:
+                self._wait_buffer("gout", cmd.in_buffer)
+                self._gout_bufs[cmd.in_buffer].copy_(d)
+            return d
 
     def _backward_and_reduce(self, cmd):
         """Install DP hooks, run backward, reset hooks — the one
@@ -428,6 +488,10 @@ class Worker:
         layer's grads final the moment that layer's backward completes
         — its bucket's all-reduce then overlaps the remaining layers'
         dgrad/wgrad kernels, same as the eager path."""
+        # the AllReduce backward is the batch's synchronization point:
+        # join the µbatch side streams (its flush reads every µbatch's
+        # deferred chunks) and run on the MAIN stream.
+        self._mu_join()
         if getattr(self, "_defer_active", False):
             self.model._flush_in_backward = True
         if self.reducer is not None:
@@ -436,7 +500,7 @@ class Worker:
             self.model.register_post_grad_hook(
                 lambda params: self.reducer.finalize())
         try:
-            self._backward_acc(cmd)
+            self._backward_acc(cmd, _mu_stream_ok=False)
         finally:
             self.model._flush_in_backward = False
             if self.reducer is not None:

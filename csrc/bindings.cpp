@@ -38,6 +38,8 @@ void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
 void ss_row_argmax(const void*, void*, int, int, hipStream_t);
 }
 // C++-linkage (gemm256.hip / wgrad256.hip)
+bool ss_gemm_nt_256w(const void*, const void*, const void*, void*, int, int,
+                     int, bool, hipStream_t);
 bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
                     int, bool, hipStream_t);
 bool ss_wgrad_tn_256(const void*, const void*, void*, int, int, int,
@@ -109,6 +111,24 @@ torch::Tensor gemm_nt_256(torch::Tensor a, torch::Tensor b,
     TORCH_CHECK(ss_gemm_nt_256(a.data_ptr(), b.data_ptr(), bias_p,
                                c.data_ptr(), M, N, K, relu, cur_stream()),
                 "shape outside the 256-tile tier: ", M, "x", N, "x", K);
+    return c;
+}
+
+torch::Tensor gemm_nt_256w(torch::Tensor a, torch::Tensor b,
+                           torch::Tensor bias, bool relu) {
+    check_bf16(a, "a");
+    check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "K mismatch");
+    const void* bias_p = nullptr;
+    if (has(bias)) {
+        check_bf16(bias, "bias");
+        bias_p = bias.data_ptr();
+    }
+    auto c = torch::empty({M, N}, a.options());
+    TORCH_CHECK(ss_gemm_nt_256w(a.data_ptr(), b.data_ptr(), bias_p,
+                                c.data_ptr(), M, N, K, relu, cur_stream()),
+                "shape outside the 256w tier: ", M, "x", N, "x", K);
     return c;
 }
 
@@ -390,6 +410,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_nt", &gemm_nt, "C = A @ B^T (+bias)(+relu) with optional A-mask");
     m.def("gemm_nt_256", &gemm_nt_256,
           "256-tile 8-phase GEMM (M%256==N%256==K%128==0)");
+    m.def("gemm_nt_256w", &gemm_nt_256w,
+          "256-tile 8-phase GEMM, 4-wave 128x128-wave-tile variant");
     m.def("wgrad_tn_256", &wgrad_tn_256,
           "256-tile 8-phase wgrad: gw += dy^T @ x (no mask/bias)");
     m.def("fp8_quantize", &fp8_quantize,

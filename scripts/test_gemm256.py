@@ -35,40 +35,72 @@ def ref(a, b, bias=None, relu=False):
     return c
 
 
-def check(M, N, K, bias_on, relu, seed=0):
+def check(M, N, K, bias_on, relu, seed=0, fn=None, tag=""):
+    fn = fn or e.gemm_nt_256
     g = torch.Generator(device="cuda").manual_seed(seed)
     a = (torch.rand((M, K), generator=g, device=dev) * 2 - 1).bfloat16()
     b = (torch.rand((N, K), generator=g, device=dev) * 2 - 1).bfloat16()
     bias = ((torch.rand((N,), generator=g, device=dev) * 2 - 1).bfloat16()
             if bias_on else EMPTY)
-    c = e.gemm_nt_256(a, b, bias, relu)
+    c = fn(a, b, bias, relu)
     want = ref(a, b, bias if bias_on else None, relu)
     err = (c.float() - want).abs()
     scale = want.abs().clamp_min(1.0)
     rel = (err / scale).max().item()
     ok = rel < 0.02
-    print(f"  {M}x{N}x{K} bias={bias_on} relu={relu}: "
+    print(f"  {tag}{M}x{N}x{K} bias={bias_on} relu={relu}: "
           f"max_rel={rel:.4f} {'OK' if ok else 'FAIL'}")
     return ok
 
 
-def race_screen(M, N, K, runs=10):
+def race_screen(M, N, K, runs=10, fn=None, tag=""):
+    fn = fn or e.gemm_nt_256
     g = torch.Generator(device="cuda").manual_seed(7)
     a = (torch.rand((M, K), generator=g, device=dev) * 2 - 1).bfloat16()
     b = (torch.rand((N, K), generator=g, device=dev) * 2 - 1).bfloat16()
-    first = e.gemm_nt_256(a, b, EMPTY, False).clone()
+    first = fn(a, b, EMPTY, False).clone()
     ok = True
     for r in range(runs - 1):
         # vary surrounding stream contents to shake scheduling
         _ = torch.randn(512 * (r + 1), 512, device=dev) @ \
             torch.randn(512, 64, device=dev)
-        c = e.gemm_nt_256(a, b, EMPTY, False)
+        c = fn(a, b, EMPTY, False)
         if not torch.equal(c, first):
             nbad = (c != first).sum().item()
-            print(f"  RACE at {M}x{N}x{K} run {r}: {nbad} mismatches")
+            print(f"  RACE at {tag}{M}x{N}x{K} run {r}: {nbad} mismatches")
             ok = False
     torch.cuda.synchronize()
-    print(f"  race screen {M}x{N}x{K} x{runs}: {'OK' if ok else 'FAIL'}")
+    print(f"  race screen {tag}{M}x{N}x{K} x{runs}: {'OK' if ok else 'FAIL'}")
+    return ok
+
+
+def v2_suite():
+    """refcheck + race + perf for the 4-wave 128x128-wave-tile variant
+    (gemm256w.hip) vs the shipped 8-wave kernel and the library."""
+    ok = True
+    print("== v2 (4-wave) refcheck ==")
+    for args in [(256, 256, 128, False, False), (256, 256, 256, True, False),
+                 (512, 256, 384, True, True), (256, 512, 512, False, True),
+                 (512, 512, 1024, True, False), (768, 256, 2048, True, True),
+                 (4096, 4096, 4096, True, False)]:
+        ok &= check(*args, fn=e.gemm_nt_256w, tag="v2 ")
+    print("== v2 race screen ==")
+    ok &= race_screen(512, 512, 1024, fn=e.gemm_nt_256w, tag="v2 ")
+    ok &= race_screen(4096, 4096, 4096, runs=6, fn=e.gemm_nt_256w, tag="v2 ")
+    print("== v2 perf (v1 | v2 | lib) ==")
+    for (M, N, K) in [(4096, 4096, 4096), (8192, 4096, 4096),
+                      (16384, 4096, 4096), (16384, 1024, 1024),
+                      (16384, 8192, 8192)]:
+        a = torch.randn(M, K, device=dev).bfloat16()
+        b = torch.randn(N, K, device=dev).bfloat16()
+        bt = b.t().contiguous().t()
+        fl = 2.0 * M * N * K
+        t1 = bench_fn(lambda: e.gemm_nt_256(a, b, EMPTY, False))
+        t2 = bench_fn(lambda: e.gemm_nt_256w(a, b, EMPTY, False))
+        tl = bench_fn(lambda: a @ bt)
+        print(f"  {M}x{N}x{K}: v1 {fl/t1/1e12:7.1f}TF | "
+              f"v2 {fl/t2/1e12:7.1f}TF | lib {fl/tl/1e12:7.1f}TF "
+              f"(v2/lib {fl/t2/1e12/(fl/tl/1e12)*100:.0f}%)")
     return ok
 
 
@@ -150,9 +182,15 @@ def wg_perf(Kb, Mo, N):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--perf-only", action="store_true")
+    ap.add_argument("--v2-only", action="store_true",
+                    help="only the 4-wave variant suite")
     args = ap.parse_args()
 
     ok = True
+    if args.v2_only:
+        ok = v2_suite()
+        print("V2 ALL OK" if ok else "V2 FAILURES", flush=True)
+        sys.exit(0 if ok else 1)
     if not args.perf_only:
         print("== refcheck ==")
         ok &= check(256, 256, 128, False, False)

@@ -75,6 +75,7 @@ def build_ext(verbose=True):
         os.path.join(src_dir, "bindings.cpp"),
         os.path.join(src_dir, "gemm.hip"),
         os.path.join(src_dir, "gemm256.hip"),
+        os.path.join(src_dir, "gemm256w.hip"),
         os.path.join(src_dir, "wgrad256.hip"),
         os.path.join(src_dir, "fp8.hip"),
         os.path.join(src_dir, "elementwise.hip"),

@@ -18,6 +18,7 @@ overlaps backward kernels.
 """
 
 import contextlib
+import os
 
 import torch
 
@@ -144,15 +145,19 @@ class Worker:
         )
         # µbatch-parallel HIP streams (single-stage schedules only):
         # on one stage, different µbatches' forwards/backwards are
-        # data-independent (grads accumulate via f32 atomics), but each
-        # small per-µbatch kernel pays a fixed wave-ramp/launch floor —
-        # the µbatch tax.  Round-robining µbatches over 2 side streams
-        # overlaps those ramps (the MI355X-idiomatic fix: fill the
-        # 256-CU chip from concurrent streams instead of serializing
-        # small kernels).  Gated off for pp>1 (stage p2p orders the
-        # pipeline) and for the AllReduce backward (joins first).
+        # data-independent (grads accumulate via f32 atomics), so their
+        # per-kernel wave-ramp floors could in principle overlap on
+        # side streams.  MEASURED OFF by default: a raw 2-stream
+        # microbench wins (scripts/microbench_streams.py: µ4 forward
+        # 133→98 µs), but the full Worker path LOSES on hardware
+        # (GPipe µ4 0.355→0.431 ms, 1F1B µ8 0.60→0.96 ms — stream
+        # contention on chip-filling kernels, same conclusion as the
+        # round-1 side-stream wgrad experiment).  The machinery stays
+        # behind SSPEED_MU_STREAMS=1 (correctness is covered by the
+        # gloo GPU tests, which ran green with it forced on).
         self._mu_par = (
-            self.device.type == "cuda"
+            os.environ.get("SSPEED_MU_STREAMS", "0") == "1"
+            and self.device.type == "cuda"
             and self.topo.pp == 1
             and getattr(schedule, "is_training", True)
             and schedule.num_micro_batches > 1

@@ -113,11 +113,35 @@ def linear_dgrad(dy, w, w_t=None, mask_src=None):
     if _is_gpu(dy):
         ext = _ext_for(dy)
         assert w_t is not None, "GPU dgrad needs the transposed weight copy"
+        # Tiered dispatch: a PLAIN wide dgrad (no mask, no epilogue) is
+        # exactly the "plain library GEMM" case where hipBLASLt belongs
+        # — measured 1362-1665 TF vs our 256-tile kernel's 1155-1305 at
+        # >=1024-wide shapes (scripts/test_gemm256.py ladder); at the
+        # flagship 256-wide shapes our kernel wins and keeps the lane.
+        # Fused ops (bias/ReLU epilogues, mask prologue, wgrad's f32
+        # atomic accumulate) always stay on the HIP kernels.
+        M, O = dy.shape
+        if (mask_src is None and not _DETERMINISTIC
+                and M >= 2048 and O >= 1024 and w.shape[1] >= 1024
+                and _dgrad_lib_enabled()):
+            return torch.matmul(dy, w)
         empty = torch.Tensor()
         return ext.gemm_nt(dy, w_t, empty, mask_src if mask_src is not None else empty, False)
     if mask_src is not None:
         dy = dy * (mask_src > 0).to(dy.dtype)
     return dy @ w
+
+
+_DGRAD_LIB = None
+
+
+def _dgrad_lib_enabled() -> bool:
+    global _DGRAD_LIB
+    if _DGRAD_LIB is None:
+        import os
+
+        _DGRAD_LIB = os.environ.get("SS_DGRAD_LIB", "1") == "1"
+    return _DGRAD_LIB
 
 
 def linear_wgrad_acc(dy, x, grad_w, grad_b=None, mask_src=None, split_k=0):

@@ -25,6 +25,8 @@ void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
                      hipStream_t);
 void ss_head_xent_bwd(const void*, const void*, void*, long, float,
                       hipStream_t);
+void ss_adamw_multi(const void*, int, long, float, float, float, float,
+                    float, float, float, hipStream_t);
 void ss_sgd_multi(const void*, int, long, float, float, float,
                   hipStream_t);
 void ss_ln_fwd(const void*, const void*, const void*, void*, void*, void*,
@@ -404,6 +406,19 @@ void sgd_multi(torch::Tensor desc, double lr, int64_t total,
                  (float)momentum, (float)weight_decay, cur_stream());
 }
 
+void adamw_multi(torch::Tensor desc, double lr, int64_t total, double beta1,
+                 double beta2, double eps, double weight_decay,
+                 double inv_bc1, double inv_bc2) {
+    TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
+                    desc.is_contiguous() && desc.dim() == 2 &&
+                    desc.size(1) == 9,
+                "desc must be CUDA int64 [T,9]");
+    ss_adamw_multi(desc.data_ptr(), (int)desc.size(0), (long)total,
+                   (float)lr, (float)beta1, (float)beta2, (float)eps,
+                   (float)weight_decay, (float)inv_bc1, (float)inv_bc2,
+                   cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -431,6 +446,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_mse_bwd", &head_mse_bwd);
     m.def("head_xent_bwd", &head_xent_bwd);
     m.def("sgd_multi", &sgd_multi);
+    m.def("adamw_multi", &adamw_multi);
     m.def("ln_fwd", &ln_fwd);
     m.def("ln_bwd_dx", &ln_bwd_dx);
     m.def("ln_bwd_dparam", &ln_bwd_dparam);

@@ -287,6 +287,60 @@ void ss_row_argmax(const void* x, void* out, int B, int C, hipStream_t st) {
                        (const __bf16*)x, (int*)out, B, C);
 }
 
+__global__ __launch_bounds__(256) void adamw_multi_kernel(
+    const long* __restrict__ desc, int ntensors, long total, float lr,
+    float beta1, float beta2, float eps, float weight_decay,
+    float inv_bc1, float inv_bc2) {
+    // desc row (9 longs): [master, grad, lp, lp_t, numel, cols, start,
+    // exp_avg, exp_avg_sq].  Decoupled weight decay (AdamW); bias
+    // corrections inv_bc1 = 1/(1-beta1^t), inv_bc2 = 1/(1-beta2^t)
+    // precomputed on the host.  Re-emits the bf16 compute copy and
+    // the transposed bf16 copy, same as sgd_multi_kernel.
+    long e = (long)blockIdx.x * 256 + threadIdx.x;
+    const long stride = (long)gridDim.x * 256;
+    for (; e < total; e += stride) {
+        int lo = 0, hi = ntensors - 1;
+        while (lo < hi) {
+            const int mid = (lo + hi + 1) >> 1;
+            if (desc[mid * 9 + 6] <= e) lo = mid;
+            else hi = mid - 1;
+        }
+        const long* d = desc + lo * 9;
+        const long i = e - d[6];
+        float* master = (float*)d[0];
+        const float g = ((const float*)d[1])[i];
+        float* ma = (float*)d[7];
+        float* va = (float*)d[8];
+        const float m1 = beta1 * ma[i] + (1.f - beta1) * g;
+        const float v1 = beta2 * va[i] + (1.f - beta2) * g * g;
+        ma[i] = m1;
+        va[i] = v1;
+        float p = master[i];
+        if (weight_decay != 0.f) p -= lr * weight_decay * p;
+        const float mhat = m1 * inv_bc1;
+        const float vhat = v1 * inv_bc2;
+        p -= lr * mhat / (sqrtf(vhat) + eps);
+        master[i] = p;
+        const __bf16 v = f2bf(p);
+        if (d[2]) ((__bf16*)d[2])[i] = v;
+        if (d[3]) {
+            const long cols = d[5];
+            const long rows = d[4] / cols;
+            ((__bf16*)d[3])[(i % cols) * rows + i / cols] = v;
+        }
+    }
+}
+
+void ss_adamw_multi(const void* desc, int ntensors, long total, float lr,
+                    float beta1, float beta2, float eps, float weight_decay,
+                    float inv_bc1, float inv_bc2, hipStream_t st) {
+    long blocks = (total + 255) / 256;
+    if (blocks > 1024) blocks = 1024;
+    hipLaunchKernelGGL(adamw_multi_kernel, dim3((int)blocks), dim3(256), 0,
+                       st, (const long*)desc, ntensors, total, lr, beta1,
+                       beta2, eps, weight_decay, inv_bc1, inv_bc2);
+}
+
 void ss_sgd_multi(const void* desc, int ntensors, long total, float lr,
                   float momentum, float weight_decay, hipStream_t st) {
     long blocks = (total + 255) / 256;

@@ -33,13 +33,11 @@ def save_checkpoint(path, model, topo, step: int = 0, extra=None,
             "stage_id": topo.stage_id,
             "step": step,
         }
-        if optimizer is not None and getattr(optimizer, "_vel", None) is not None:
-            state["opt"] = {
-                "velocity": [v.detach().cpu() for v in optimizer._vel],
-                "lr": optimizer.lr,
-                "momentum": optimizer.momentum,
-                "weight_decay": optimizer.weight_decay,
-            }
+        if optimizer is not None and hasattr(optimizer, "state_dict"):
+            opt_state = optimizer.state_dict()
+            # stateless SGD (no momentum): nothing worth persisting
+            if any(k in opt_state for k in ("velocity", "exp_avg")):
+                state["opt"] = opt_state
         torch.save(state, os.path.join(path, f"stage_{topo.stage_id:02d}.pt"))
     if topo.rank == 0:
         meta = {"dp": topo.dp, "pp": topo.pp, "step": step}
@@ -69,16 +67,5 @@ def load_checkpoint(path, model, topo, optimizer=None):
         p.data.copy_(saved.to(p.data.device))
         p.sync_lp()
     if optimizer is not None and "opt" in state:
-        opt_state = state["opt"]
-        vel = opt_state["velocity"]
-        if optimizer._vel is None:
-            # optimizer constructed with momentum=0 but checkpoint has
-            # velocities: adopt the saved hyperparameters
-            optimizer.momentum = opt_state["momentum"]
-            optimizer._vel = [torch.zeros_like(p.data)
-                              for p in optimizer.params]
-        assert len(vel) == len(optimizer._vel), (len(vel), len(optimizer._vel))
-        for dst, src in zip(optimizer._vel, vel):
-            assert dst.shape == src.shape, (dst.shape, src.shape)
-            dst.copy_(src.to(dst.device))
+        optimizer.load_state_dict(state["opt"])
     return meta

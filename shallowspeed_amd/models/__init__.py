@@ -12,4 +12,4 @@ from .layers import (  # noqa: F401
     SoftmaxMSE,
     SoftmaxXent,
 )
-from .optimizer import SGD  # noqa: F401
+from .optimizer import SGD, AdamW  # noqa: F401

@@ -1,11 +1,20 @@
-"""SGD — stateless by default (reference: shallowspeed/optimizer.py:4-13),
-with optional momentum and weight decay (beyond-reference extension).
+"""Optimizers.
 
-CPU: elementwise torch updates per parameter.
-GPU: ONE fused multi-tensor HIP kernel launch updates every parameter
-of the stage: f32 master -= lr * (grad + wd*master [+ momentum]), and
-re-emits the bf16 compute copy AND the transposed bf16 copy in the same
-pass (the transposed copy is what makes dgrad an NT GEMM).
+SGD — stateless by default (reference: shallowspeed/optimizer.py:4-13),
+with optional momentum and weight decay (beyond-reference extension).
+AdamW — beyond-reference: decoupled-weight-decay Adam for modern
+training recipes.
+
+CPU: elementwise torch updates per parameter (the numerics oracle).
+GPU: ONE fused multi-tensor HIP kernel launch per step updates every
+parameter of the stage — f32 master update + optimizer state, and
+re-emits the bf16 compute copy AND the transposed bf16 copy in the
+same pass (the transposed copy is what makes dgrad the same NT MFMA
+kernel as forward).
+
+Both expose state_dict()/load_state_dict() so checkpoints carry the
+full optimization trajectory (momentum velocities / Adam moments and
+step count).
 """
 
 import torch
@@ -27,7 +36,7 @@ class SGD:
 
     def _build_desc(self):
         # Static descriptor: [master_ptr, grad_ptr, lp_ptr, lpt_ptr,
-        # numel, cols, start] per tensor, int64, lives on device.
+        # numel, cols, start, vel_ptr] per tensor, int64, on device.
         # Pointers are stable because parameter storage is persistent.
         rows, start = [], 0
         for i, p in enumerate(self.params):
@@ -64,3 +73,107 @@ class SGD:
                     self._vel[i].mul_(self.momentum).add_(g)
                     g = self._vel[i]
                 p.data -= self.lr * g
+
+    # ------------------------------------------------- checkpoint state
+    def state_dict(self):
+        d = {"kind": "sgd", "lr": self.lr, "momentum": self.momentum,
+             "weight_decay": self.weight_decay}
+        if self._vel is not None:
+            d["velocity"] = [v.detach().cpu() for v in self._vel]
+        return d
+
+    def load_state_dict(self, d):
+        assert d.get("kind", "sgd") == "sgd", d.get("kind")
+        self.lr = d.get("lr", self.lr)
+        self.weight_decay = d.get("weight_decay", self.weight_decay)
+        if "velocity" in d:
+            self.momentum = d.get("momentum", self.momentum)
+            if self._vel is None:
+                self._vel = [torch.zeros_like(p.data) for p in self.params]
+                self._desc = None  # descriptor must pick up vel pointers
+            assert len(d["velocity"]) == len(self._vel)
+            for dst, src in zip(self._vel, d["velocity"]):
+                assert dst.shape == src.shape, (dst.shape, src.shape)
+                dst.copy_(src.to(dst.device))
+
+
+class AdamW:
+    """Decoupled-weight-decay Adam (beyond-reference).  GPU: one fused
+    multi-tensor HIP launch (adamw_multi_kernel) updates masters +
+    moments and re-emits bf16/bf16ᵀ copies; CPU path is the oracle
+    (matches torch.optim.AdamW semantics)."""
+
+    def __init__(self, parameters, lr: float = 1e-3, betas=(0.9, 0.999),
+                 eps: float = 1e-8, weight_decay: float = 0.01):
+        self.params = [p for p in parameters if p.requires_grad]
+        self.lr = float(lr)
+        self.betas = (float(betas[0]), float(betas[1]))
+        self.eps = float(eps)
+        self.weight_decay = float(weight_decay)
+        self.step_count = 0
+        self._m = [torch.zeros_like(p.data) for p in self.params]
+        self._v = [torch.zeros_like(p.data) for p in self.params]
+        self._desc = None
+
+    def _build_desc(self):
+        rows, start = [], 0
+        for i, p in enumerate(self.params):
+            t = p.data
+            cols = t.shape[1] if t.dim() == 2 else 1
+            rows.append([
+                t.data_ptr(), p.grad.data_ptr(),
+                p.lp.data_ptr() if p.lp is not None else 0,
+                p.lp_t.data_ptr() if p.lp_t is not None else 0,
+                t.numel(), cols, start,
+                self._m[i].data_ptr(), self._v[i].data_ptr(),
+            ])
+            start += t.numel()
+        self._total = start
+        cpu = torch.tensor(rows, dtype=torch.int64)
+        self._desc = cpu.to(self.params[0].data.device)
+
+    def step(self):
+        if not self.params:
+            return
+        self.step_count += 1
+        b1, b2 = self.betas
+        inv_bc1 = 1.0 / (1.0 - b1 ** self.step_count)
+        inv_bc2 = 1.0 / (1.0 - b2 ** self.step_count)
+        if self.params[0].data.is_cuda:
+            ext = load_ext(required=True)
+            if self._desc is None:
+                self._build_desc()
+            ext.adamw_multi(self._desc, self.lr, self._total, b1, b2,
+                            self.eps, self.weight_decay, inv_bc1, inv_bc2)
+        else:
+            for i, p in enumerate(self.params):
+                g = p.grad
+                self._m[i].mul_(b1).add_(g, alpha=1 - b1)
+                self._v[i].mul_(b2).addcmul_(g, g, value=1 - b2)
+                if self.weight_decay:
+                    p.data.mul_(1 - self.lr * self.weight_decay)
+                mhat = self._m[i] * inv_bc1
+                vhat = self._v[i] * inv_bc2
+                p.data -= self.lr * mhat / (vhat.sqrt() + self.eps)
+
+    # ------------------------------------------------- checkpoint state
+    def state_dict(self):
+        return {
+            "kind": "adamw", "lr": self.lr, "betas": self.betas,
+            "eps": self.eps, "weight_decay": self.weight_decay,
+            "step": self.step_count,
+            "exp_avg": [m.detach().cpu() for m in self._m],
+            "exp_avg_sq": [v.detach().cpu() for v in self._v],
+        }
+
+    def load_state_dict(self, d):
+        assert d.get("kind") == "adamw", d.get("kind")
+        self.lr = d.get("lr", self.lr)
+        self.betas = tuple(d.get("betas", self.betas))
+        self.eps = d.get("eps", self.eps)
+        self.weight_decay = d.get("weight_decay", self.weight_decay)
+        self.step_count = d.get("step", 0)
+        for dst, src in zip(self._m, d["exp_avg"]):
+            dst.copy_(src.to(dst.device))
+        for dst, src in zip(self._v, d["exp_avg_sq"]):
+            dst.copy_(src.to(dst.device))

@@ -72,6 +72,9 @@ def main():
     ap.add_argument("--lr", type=float, default=0.006)
     ap.add_argument("--momentum", type=float, default=0.0)
     ap.add_argument("--weight-decay", type=float, default=0.0)
+    ap.add_argument("--optimizer", choices=["sgd", "adamw"], default="sgd",
+                    help="sgd matches the reference (optimizer.py:4-13); "
+                         "adamw is the fused multi-tensor AdamW extension")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise run-to-run reproducible GPU training "
                          "(forces the single-owner split_k=1 wgrad path; "
@@ -120,8 +123,15 @@ def main():
     model = MLP(args.layer_sizes, stage_idx=topo.stage_id, n_stages=args.pp,
                 global_batch_size=args.global_batch, loss=args.loss)
     model.materialize_device(device)
-    optimizer = SGD(model.parameters(), lr=args.lr, momentum=args.momentum,
-                    weight_decay=args.weight_decay)
+    if args.optimizer == "adamw":
+        from shallowspeed_amd.models import AdamW
+
+        optimizer = AdamW(model.parameters(), lr=args.lr,
+                          weight_decay=args.weight_decay)
+    else:
+        optimizer = SGD(model.parameters(), lr=args.lr,
+                        momentum=args.momentum,
+                        weight_decay=args.weight_decay)
 
     if args.resume:
         from shallowspeed_amd.checkpoint import load_checkpoint

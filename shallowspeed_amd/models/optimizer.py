@@ -22,13 +22,27 @@ import torch
 from ..ops._ext import load_ext
 
 
+def _clip_flat(flat_grad, clip_norm):
+    """Global-norm gradient clip on the stage's flat f32 grad buffer.
+    Runs AFTER the DP all-reduce (grads are the summed global grads),
+    so every replica computes the identical scale — sync-free and
+    deterministic.  Returns the pre-clip norm (a 0-d tensor; never
+    host-synced on the hot path)."""
+    n = torch.linalg.vector_norm(flat_grad)
+    scale = (clip_norm / (n + 1e-6)).clamp(max=1.0)
+    flat_grad.mul_(scale)
+    return n
+
+
 class SGD:
     def __init__(self, parameters, lr: float, momentum: float = 0.0,
-                 weight_decay: float = 0.0):
+                 weight_decay: float = 0.0, clip_norm=None, flat_grad=None):
         self.params = [p for p in parameters if p.requires_grad]
         self.lr = float(lr)
         self.momentum = float(momentum)
         self.weight_decay = float(weight_decay)
+        self.clip_norm = clip_norm
+        self.flat_grad = flat_grad  # model._flat_grad (clipping substrate)
         self._vel = None
         if self.momentum:
             self._vel = [torch.zeros_like(p.data) for p in self.params]
@@ -58,6 +72,8 @@ class SGD:
     def step(self):
         if not self.params:
             return
+        if self.clip_norm is not None and self.flat_grad is not None:
+            _clip_flat(self.flat_grad, self.clip_norm)
         if self.params[0].data.is_cuda:
             ext = load_ext(required=True)
             if self._desc is None:
@@ -104,12 +120,15 @@ class AdamW:
     (matches torch.optim.AdamW semantics)."""
 
     def __init__(self, parameters, lr: float = 1e-3, betas=(0.9, 0.999),
-                 eps: float = 1e-8, weight_decay: float = 0.01):
+                 eps: float = 1e-8, weight_decay: float = 0.01,
+                 clip_norm=None, flat_grad=None):
         self.params = [p for p in parameters if p.requires_grad]
         self.lr = float(lr)
         self.betas = (float(betas[0]), float(betas[1]))
         self.eps = float(eps)
         self.weight_decay = float(weight_decay)
+        self.clip_norm = clip_norm
+        self.flat_grad = flat_grad
         self.step_count = 0
         self._m = [torch.zeros_like(p.data) for p in self.params]
         self._v = [torch.zeros_like(p.data) for p in self.params]
@@ -135,6 +154,8 @@ class AdamW:
     def step(self):
         if not self.params:
             return
+        if self.clip_norm is not None and self.flat_grad is not None:
+            _clip_flat(self.flat_grad, self.clip_norm)
         self.step_count += 1
         b1, b2 = self.betas
         inv_bc1 = 1.0 / (1.0 - b1 ** self.step_count)

@@ -88,3 +88,23 @@ def test_adamw_fused_kernel_matches_cpu(gpu_device):
         if gp.lp_t is not None:
             torch.testing.assert_close(gp.lp_t.cpu(), gp.lp.t().cpu(),
                                        rtol=0, atol=0)
+
+
+def test_grad_clip_global_norm():
+    from shallowspeed_amd.models import SGD
+
+    model = MLP(SIZES, 0, 1, 16).materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=0.0, clip_norm=1.0,
+              flat_grad=model._flat_grad)
+    for p, g in zip(model.parameters(), _grads_like(model, 1)):
+        p.grad.copy_(g * 10)  # force norm >> 1
+    pre = model._flat_grad.norm().item()
+    assert pre > 1.0
+    opt.step()
+    post = model._flat_grad.norm().item()
+    assert abs(post - 1.0) < 1e-4, post
+    # below-threshold grads are untouched
+    model._flat_grad.mul_(0.3)
+    want = model._flat_grad.clone()
+    opt.step()
+    torch.testing.assert_close(model._flat_grad, want, rtol=1e-6, atol=0)

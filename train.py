@@ -75,6 +75,10 @@ def main():
     ap.add_argument("--optimizer", choices=["sgd", "adamw"], default="sgd",
                     help="sgd matches the reference (optimizer.py:4-13); "
                          "adamw is the fused multi-tensor AdamW extension")
+    ap.add_argument("--clip-grad", type=float, default=None,
+                    help="global-norm gradient clipping threshold "
+                         "(applied on the flat grad buffer after the DP "
+                         "all-reduce, before the optimizer step)")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise run-to-run reproducible GPU training "
                          "(forces the single-owner split_k=1 wgrad path; "
@@ -123,15 +127,16 @@ def main():
     model = MLP(args.layer_sizes, stage_idx=topo.stage_id, n_stages=args.pp,
                 global_batch_size=args.global_batch, loss=args.loss)
     model.materialize_device(device)
+    clip_kw = dict(clip_norm=args.clip_grad, flat_grad=model._flat_grad)
     if args.optimizer == "adamw":
         from shallowspeed_amd.models import AdamW
 
         optimizer = AdamW(model.parameters(), lr=args.lr,
-                          weight_decay=args.weight_decay)
+                          weight_decay=args.weight_decay, **clip_kw)
     else:
         optimizer = SGD(model.parameters(), lr=args.lr,
                         momentum=args.momentum,
-                        weight_decay=args.weight_decay)
+                        weight_decay=args.weight_decay, **clip_kw)
 
     if args.resume:
         from shallowspeed_amd.checkpoint import load_checkpoint

@@ -25,6 +25,7 @@ void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
                      hipStream_t);
 void ss_head_xent_bwd(const void*, const void*, void*, long, float,
                       hipStream_t);
+void ss_transpose_bf16(const void*, void*, int, int, hipStream_t);
 void ss_adamw_multi(const void*, int, long, float, float, float, float,
                     float, float, float, hipStream_t);
 void ss_sgd_multi(const void*, int, long, float, float, float,
@@ -419,6 +420,18 @@ void adamw_multi(torch::Tensor desc, double lr, int64_t total, double beta1,
                    cur_stream());
 }
 
+void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
+    check_bf16(src, "src");
+    check_bf16(dst, "dst");
+    const int rows = src.size(0), cols = src.size(1);
+    TORCH_CHECK(src.dim() == 2 && dst.dim() == 2 &&
+                    dst.size(0) == cols && dst.size(1) == rows &&
+                    rows % 64 == 0 && cols % 64 == 0,
+                "transpose_bf16 needs 64-aligned 2-D shapes");
+    ss_transpose_bf16(src.data_ptr(), dst.data_ptr(), rows, cols,
+                      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -447,6 +460,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_xent_bwd", &head_xent_bwd);
     m.def("sgd_multi", &sgd_multi);
     m.def("adamw_multi", &adamw_multi);
+    m.def("transpose_bf16", &transpose_bf16);
     m.def("ln_fwd", &ln_fwd);
     m.def("ln_bwd_dx", &ln_bwd_dx);
     m.def("ln_bwd_dparam", &ln_bwd_dparam);

@@ -287,6 +287,44 @@ void ss_row_argmax(const void* x, void* out, int B, int C, hipStream_t st) {
                        (const __bf16*)x, (int*)out, B, C);
 }
 
+// Tiled bf16 transpose: dst[c][r] = src[r][c], 64x64 LDS tiles with
+// padding — replaces the fused optimizers' in-kernel transposed
+// emission for LARGE 2-D weights, where the per-element scatter store
+// (stride rows*2 B) is uncoalesced and dominated the wide-model step
+// (measured: sgd_multi 1.65 ms/step on 8x4096-wide — ~7x the
+// coalesced roofline; see profiles/r02_evidence.md).
+__global__ __launch_bounds__(256) void transpose_bf16_kernel(
+    const __bf16* __restrict__ src, __bf16* __restrict__ dst, int rows,
+    int cols) {
+    __shared__ __bf16 tile[64][65];
+    const int t = threadIdx.x;
+    const int r0 = blockIdx.y * 64, c0 = blockIdx.x * 64;
+    const int lr = t >> 2;           // 0..63
+    const int lc = (t & 3) * 16;     // 0,16,32,48
+    // load 64x64 source tile, 16B vectors
+    {
+        const __bf16* p = src + (long)(r0 + lr) * cols + c0 + lc;
+#pragma unroll
+        for (int v = 0; v < 2; ++v) {
+            bf16x8 x = *(const bf16x8*)(p + v * 8);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) tile[lr][lc + v * 8 + e] = x[e];
+        }
+    }
+    __syncthreads();
+    // write transposed, 16B vectors along dst rows (= source columns)
+    {
+        __bf16* p = dst + (long)(c0 + lr) * rows + r0 + lc;
+#pragma unroll
+        for (int v = 0; v < 2; ++v) {
+            bf16x8 x;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) x[e] = tile[lc + v * 8 + e][lr];
+            *(bf16x8*)(p + v * 8) = x;
+        }
+    }
+}
+
 __global__ __launch_bounds__(256) void adamw_multi_kernel(
     const long* __restrict__ desc, int ntensors, long total, float lr,
     float beta1, float beta2, float eps, float weight_decay,
@@ -329,6 +367,13 @@ __global__ __launch_bounds__(256) void adamw_multi_kernel(
             ((__bf16*)d[3])[(i % cols) * rows + i / cols] = v;
         }
     }
+}
+
+void ss_transpose_bf16(const void* src, void* dst, int rows, int cols,
+                       hipStream_t st) {
+    dim3 grid(cols / 64, rows / 64);
+    hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, st,
+                       (const __bf16*)src, (__bf16*)dst, rows, cols);
 }
 
 void ss_adamw_multi(const void* desc, int ntensors, long total, float lr,

@@ -646,6 +646,37 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 
 // ---------------------------------------------------------------- colsum
 
+// 8-wide vectorized variant: each thread owns 8 contiguous columns
+// (one bf16x8 load per row), 8x the bytes in flight per thread vs the
+// scalar kernel (measured 195 us at 16384x4096 = 11x off roofline —
+// latency-bound scalar loads).
+template <bool HAS_MASK>
+__global__ __launch_bounds__(256) void colsum8_kernel(
+    const __bf16* __restrict__ dY,    // [M][N]
+    const __bf16* __restrict__ mask,  // [M][N]
+    float* __restrict__ gb,           // [N] (atomicAdd +=)
+    int M, int N, int rows_per_block) {
+    const int col0 = (blockIdx.x * 256 + threadIdx.x) * 8;
+    const int r0 = blockIdx.y * rows_per_block;
+    const int r1 = min(M, r0 + rows_per_block);
+    if (col0 >= N) return;
+    float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    for (int r = r0; r < r1; ++r) {
+        bf16x8 v = *(const bf16x8*)(dY + (long)r * N + col0);
+        if constexpr (HAS_MASK) {
+            bf16x8 m = *(const bf16x8*)(mask + (long)r * N + col0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                if (bf2f(m[e]) > 0.f) s[e] += bf2f(v[e]);
+        } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) s[e] += bf2f(v[e]);
+        }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) atomicAdd(&gb[col0 + e], s[e]);
+}
+
 template <bool HAS_MASK>
 __global__ __launch_bounds__(256) void colsum_kernel(
     const __bf16* __restrict__ dY,    // [M][N]
@@ -870,13 +901,28 @@ void ss_wgrad_tn_multi(const void* chunk_table, int nchunks, bool has_mask,
 void ss_colsum(const void* dY, const void* mask, void* gb, int M, int N,
                hipStream_t stream) {
     // enough blocks to cover the CUs (256 CUs; Guideline 11)
-    int col_tiles = cdiv(N, 256);
+    const bool vec8 = (N % 8 == 0) && N >= 2048;
+    const int cols_per_block = vec8 ? 2048 : 256;
+    int col_tiles = cdiv(N, cols_per_block);
     int splits = cdiv(512, col_tiles);
     if (splits > cdiv(M, 64)) splits = cdiv(M, 64);
     if (splits < 1) splits = 1;
     int rows_per_block = cdiv(M, splits);
-    dim3 grid(cdiv(N, 256), splits);
+    dim3 grid(col_tiles, splits);
     dim3 block(256);
+    if (vec8) {
+        if (mask)
+            hipLaunchKernelGGL((colsum8_kernel<true>), grid, block, 0,
+                               stream, (const __bf16*)dY,
+                               (const __bf16*)mask, (float*)gb, M, N,
+                               rows_per_block);
+        else
+            hipLaunchKernelGGL((colsum8_kernel<false>), grid, block, 0,
+                               stream, (const __bf16*)dY,
+                               (const __bf16*)mask, (float*)gb, M, N,
+                               rows_per_block);
+        return;
+    }
     if (mask)
         hipLaunchKernelGGL((colsum_kernel<true>), grid, block, 0, stream,
                            (const __bf16*)dY, (const __bf16*)mask, (float*)gb,

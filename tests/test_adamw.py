@@ -110,3 +110,38 @@ def test_grad_clip_global_norm():
     want = model._flat_grad.clone()
     opt.step()
     torch.testing.assert_close(model._flat_grad, want, rtol=1e-6, atol=0)
+
+
+@pytest.mark.gpu
+def test_transpose_bf16_tiled_kernel(gpu_device):
+    from shallowspeed_amd.ops import load_ext
+
+    e = load_ext(required=True)
+    g = torch.Generator(device="cuda").manual_seed(3)
+    x = torch.randn(2048, 1024, generator=g, device=gpu_device).bfloat16()
+    dst = torch.empty(1024, 2048, dtype=torch.bfloat16, device=gpu_device)
+    e.transpose_bf16(x, dst)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, x.t().contiguous())
+
+
+@pytest.mark.gpu
+def test_sgd_tiled_transpose_path(gpu_device):
+    """Wide weights (cols >= 1024) take the tiled-transpose emission;
+    lp_t must still equal lp.t() bitwise after a fused step."""
+    from shallowspeed_amd.models import SGD
+
+    model = MLP([1024, 1024, 10], 0, 1, 16).materialize_device(gpu_device)
+    opt = SGD(model.parameters(), lr=0.01)
+    for p, gr in zip(model.parameters(), _grads_like(model, 5)):
+        p.grad.copy_(gr.to(gpu_device))
+    opt.step()
+    torch.cuda.synchronize()
+    assert any(id(p) in {id(q) for q, _, _ in opt._tiled_t}
+               for p in model.parameters()), "tiled path not engaged"
+    for p in model.parameters():
+        if p.lp_t is not None:
+            assert torch.equal(p.lp_t, p.lp.t().contiguous())
+        torch.testing.assert_close(p.lp.float(),
+                                   p.data.to(torch.bfloat16).float(),
+                                   rtol=0, atol=0)

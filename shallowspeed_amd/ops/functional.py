@@ -337,7 +337,14 @@ def linear_wgrad_multi(chunks, grad_w, grad_b=None, split_k=0):
     if split_k == 0 and _DETERMINISTIC:
         split_k = 1
     dy0 = chunks[0][0]
-    if not _is_gpu(dy0):
+    if not _is_gpu(dy0) or split_k == 1:
+        # split_k==1 (deterministic mode): the chunked kernel would
+        # still run one block PER CHUNK per tile, all atomicAdd-ing
+        # the same gW element — chunk ARRIVAL order is scheduling-
+        # dependent, which broke bitwise reproducibility (flaky GPU
+        # determinism test).  Sequential per-chunk launches fix the
+        # accumulation order at a small launch-count cost; perf mode
+        # (split_k=0) keeps the single fused launch.
         for dy, x, m in chunks:
             linear_wgrad_acc(dy, x, grad_w, grad_b, m, split_k)
         return

@@ -26,6 +26,10 @@ void ss_head_mse_bwd(const void*, const void*, void*, int, int, float,
 void ss_head_xent_bwd(const void*, const void*, void*, long, float,
                       hipStream_t);
 void ss_transpose_bf16(const void*, void*, int, int, hipStream_t);
+void ss_sgd_multi2(const void*, const void*, int, float, float, float,
+                   hipStream_t);
+void ss_adamw_multi2(const void*, const void*, int, float, float, float,
+                     float, float, float, float, hipStream_t);
 void ss_adamw_multi(const void*, int, long, float, float, float, float,
                     float, float, float, hipStream_t);
 void ss_sgd_multi(const void*, int, long, float, float, float,
@@ -420,6 +424,38 @@ void adamw_multi(torch::Tensor desc, double lr, int64_t total, double beta1,
                    cur_stream());
 }
 
+void sgd_multi2(torch::Tensor desc, torch::Tensor bmap, double lr,
+                double momentum, double weight_decay) {
+    TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
+                    desc.is_contiguous() && desc.dim() == 2 &&
+                    desc.size(1) == 8,
+                "desc must be CUDA int64 [T,8]");
+    TORCH_CHECK(bmap.is_cuda() && bmap.scalar_type() == torch::kLong &&
+                    bmap.is_contiguous() && bmap.dim() == 2 &&
+                    bmap.size(1) == 2,
+                "bmap must be CUDA int64 [B,2]");
+    ss_sgd_multi2(desc.data_ptr(), bmap.data_ptr(), (int)bmap.size(0),
+                  (float)lr, (float)momentum, (float)weight_decay,
+                  cur_stream());
+}
+
+void adamw_multi2(torch::Tensor desc, torch::Tensor bmap, double lr,
+                  double beta1, double beta2, double eps,
+                  double weight_decay, double inv_bc1, double inv_bc2) {
+    TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
+                    desc.is_contiguous() && desc.dim() == 2 &&
+                    desc.size(1) == 9,
+                "desc must be CUDA int64 [T,9]");
+    TORCH_CHECK(bmap.is_cuda() && bmap.scalar_type() == torch::kLong &&
+                    bmap.is_contiguous() && bmap.dim() == 2 &&
+                    bmap.size(1) == 2,
+                "bmap must be CUDA int64 [B,2]");
+    ss_adamw_multi2(desc.data_ptr(), bmap.data_ptr(), (int)bmap.size(0),
+                    (float)lr, (float)beta1, (float)beta2, (float)eps,
+                    (float)weight_decay, (float)inv_bc1, (float)inv_bc2,
+                    cur_stream());
+}
+
 void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
     check_bf16(src, "src");
     check_bf16(dst, "dst");
@@ -461,6 +497,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_multi", &sgd_multi);
     m.def("adamw_multi", &adamw_multi);
     m.def("transpose_bf16", &transpose_bf16);
+    m.def("sgd_multi2", &sgd_multi2);
+    m.def("adamw_multi2", &adamw_multi2);
     m.def("ln_fwd", &ln_fwd);
     m.def("ln_bwd_dx", &ln_bwd_dx);
     m.def("ln_bwd_dparam", &ln_bwd_dparam);

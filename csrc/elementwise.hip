@@ -287,6 +287,169 @@ void ss_row_argmax(const void* x, void* out, int B, int C, hipStream_t st) {
                        (const __bf16*)x, (int*)out, B, C);
 }
 
+// ---- blockmap fused optimizers ------------------------------------
+// v2 of the fused optimizer kernels: a host-precomputed block→(tensor,
+// elem_base) map replaces the per-ELEMENT binary search of the v1
+// kernels — the dependent 6-level desc walk serialized every loop
+// iteration and held the wide-model sgd_multi at ~0.9 TB/s (8.2 ms/step
+// at 523M params; profiles/r02_kernel_stats.md).  Each block owns
+// 16384 contiguous elements of ONE tensor; float4/ushort4 vector
+// accesses; the only scalar fallback is the grad operand when its
+// flat-buffer view is not 16B-aligned (odd bias offsets).
+#define SS_OPT_EPB 16384
+
+__global__ __launch_bounds__(256) void sgd_multi2_kernel(
+    const long* __restrict__ desc, const long* __restrict__ bmap, float lr,
+    float momentum, float weight_decay) {
+    const long* d = desc + bmap[(long)blockIdx.x * 2] * 8;
+    const long base = bmap[(long)blockIdx.x * 2 + 1];
+    const long n = d[4];
+    float* __restrict__ master = (float*)d[0];
+    const float* __restrict__ grad = (const float*)d[1];
+    __bf16* lp = (__bf16*)d[2];
+    __bf16* lpt = (__bf16*)d[3];
+    float* vel = (float*)d[7];
+    const long cols = d[5];
+    const long rows = cols ? n / cols : 0;
+    const bool ga = ((unsigned long)d[1] & 15) == 0;
+#pragma unroll 1
+    for (int it = 0; it < SS_OPT_EPB / 1024; ++it) {
+        const long e = base + (long)it * 1024 + threadIdx.x * 4;
+        if (e >= n) break;
+        if (e + 4 <= n) {
+            float g4[4];
+            if (ga) {
+                const float4 gv = *(const float4*)(grad + e);
+                g4[0] = gv.x; g4[1] = gv.y; g4[2] = gv.z; g4[3] = gv.w;
+            } else {
+#pragma unroll
+                for (int k = 0; k < 4; ++k) g4[k] = grad[e + k];
+            }
+            float4 m4 = *(float4*)(master + e);
+            float* mv = (float*)&m4;
+            float v4[4];
+            if (vel) {
+                float4 vv = *(float4*)(vel + e);
+                v4[0] = vv.x; v4[1] = vv.y; v4[2] = vv.z; v4[3] = vv.w;
+            }
+            __bf16 out[4];
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+                float g = g4[k];
+                if (weight_decay != 0.f) g += weight_decay * mv[k];
+                if (vel) {
+                    g = momentum * v4[k] + g;
+                    v4[k] = g;
+                }
+                mv[k] -= lr * g;
+                out[k] = f2bf(mv[k]);
+            }
+            *(float4*)(master + e) = m4;
+            if (vel) {
+                float4 vv = {v4[0], v4[1], v4[2], v4[3]};
+                *(float4*)(vel + e) = vv;
+            }
+            if (lp) *(ushort4*)(lp + e) = *(ushort4*)out;
+            if (lpt) {
+#pragma unroll
+                for (int k = 0; k < 4; ++k)
+                    lpt[((e + k) % cols) * rows + (e + k) / cols] = out[k];
+            }
+        } else {
+            for (long q = e; q < n; ++q) {
+                float g = grad[q];
+                if (weight_decay != 0.f) g += weight_decay * master[q];
+                if (vel) {
+                    g = momentum * vel[q] + g;
+                    vel[q] = g;
+                }
+                const float m = master[q] - lr * g;
+                master[q] = m;
+                const __bf16 v = f2bf(m);
+                if (lp) lp[q] = v;
+                if (lpt) lpt[(q % cols) * rows + q / cols] = v;
+            }
+        }
+    }
+}
+
+__global__ __launch_bounds__(256) void adamw_multi2_kernel(
+    const long* __restrict__ desc, const long* __restrict__ bmap, float lr,
+    float beta1, float beta2, float eps, float weight_decay, float inv_bc1,
+    float inv_bc2) {
+    const long* d = desc + bmap[(long)blockIdx.x * 2] * 9;
+    const long base = bmap[(long)blockIdx.x * 2 + 1];
+    const long n = d[4];
+    float* __restrict__ master = (float*)d[0];
+    const float* __restrict__ grad = (const float*)d[1];
+    __bf16* lp = (__bf16*)d[2];
+    __bf16* lpt = (__bf16*)d[3];
+    float* __restrict__ ma = (float*)d[7];
+    float* __restrict__ va = (float*)d[8];
+    const long cols = d[5];
+    const long rows = cols ? n / cols : 0;
+    const bool ga = ((unsigned long)d[1] & 15) == 0;
+#pragma unroll 1
+    for (int it = 0; it < SS_OPT_EPB / 1024; ++it) {
+        const long e = base + (long)it * 1024 + threadIdx.x * 4;
+        if (e >= n) break;
+        if (e + 4 <= n) {
+            float g4[4];
+            if (ga) {
+                const float4 gv = *(const float4*)(grad + e);
+                g4[0] = gv.x; g4[1] = gv.y; g4[2] = gv.z; g4[3] = gv.w;
+            } else {
+#pragma unroll
+                for (int k = 0; k < 4; ++k) g4[k] = grad[e + k];
+            }
+            float4 m4 = *(float4*)(master + e);
+            float4 ma4 = *(float4*)(ma + e);
+            float4 va4 = *(float4*)(va + e);
+            float* mv = (float*)&m4;
+            float* mav = (float*)&ma4;
+            float* vav = (float*)&va4;
+            __bf16 out[4];
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+                const float g = g4[k];
+                const float m1 = beta1 * mav[k] + (1.f - beta1) * g;
+                const float v1 = beta2 * vav[k] + (1.f - beta2) * g * g;
+                mav[k] = m1;
+                vav[k] = v1;
+                float pv = mv[k];
+                if (weight_decay != 0.f) pv -= lr * weight_decay * pv;
+                pv -= lr * (m1 * inv_bc1) / (sqrtf(v1 * inv_bc2) + eps);
+                mv[k] = pv;
+                out[k] = f2bf(pv);
+            }
+            *(float4*)(master + e) = m4;
+            *(float4*)(ma + e) = ma4;
+            *(float4*)(va + e) = va4;
+            if (lp) *(ushort4*)(lp + e) = *(ushort4*)out;
+            if (lpt) {
+#pragma unroll
+                for (int k = 0; k < 4; ++k)
+                    lpt[((e + k) % cols) * rows + (e + k) / cols] = out[k];
+            }
+        } else {
+            for (long q = e; q < n; ++q) {
+                const float g = grad[q];
+                const float m1 = beta1 * ma[q] + (1.f - beta1) * g;
+                const float v1 = beta2 * va[q] + (1.f - beta2) * g * g;
+                ma[q] = m1;
+                va[q] = v1;
+                float pv = master[q];
+                if (weight_decay != 0.f) pv -= lr * weight_decay * pv;
+                pv -= lr * (m1 * inv_bc1) / (sqrtf(v1 * inv_bc2) + eps);
+                master[q] = pv;
+                const __bf16 v = f2bf(pv);
+                if (lp) lp[q] = v;
+                if (lpt) lpt[(q % cols) * rows + q / cols] = v;
+            }
+        }
+    }
+}
+
 // Tiled bf16 transpose: dst[c][r] = src[r][c], 64x64 LDS tiles with
 // padding — replaces the fused optimizers' in-kernel transposed
 // emission for LARGE 2-D weights, where the per-element scatter store
@@ -367,6 +530,23 @@ __global__ __launch_bounds__(256) void adamw_multi_kernel(
             ((__bf16*)d[3])[(i % cols) * rows + i / cols] = v;
         }
     }
+}
+
+void ss_sgd_multi2(const void* desc, const void* bmap, int nblocks,
+                   float lr, float momentum, float weight_decay,
+                   hipStream_t st) {
+    hipLaunchKernelGGL(sgd_multi2_kernel, dim3(nblocks), dim3(256), 0, st,
+                       (const long*)desc, (const long*)bmap, lr, momentum,
+                       weight_decay);
+}
+
+void ss_adamw_multi2(const void* desc, const void* bmap, int nblocks,
+                     float lr, float beta1, float beta2, float eps,
+                     float weight_decay, float inv_bc1, float inv_bc2,
+                     hipStream_t st) {
+    hipLaunchKernelGGL(adamw_multi2_kernel, dim3(nblocks), dim3(256), 0, st,
+                       (const long*)desc, (const long*)bmap, lr, beta1,
+                       beta2, eps, weight_decay, inv_bc1, inv_bc2);
 }
 
 void ss_transpose_bf16(const void* src, void* dst, int rows, int cols,

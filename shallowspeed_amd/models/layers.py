@@ -36,6 +36,18 @@ import torch
 from ..ops import functional as F
 
 
+_DZ_FIRST = None
+
+
+def _dz_first_enabled() -> bool:
+    global _DZ_FIRST
+    if _DZ_FIRST is None:
+        import os
+
+        _DZ_FIRST = os.environ.get("SS_DZ_FIRST", "1") == "1"
+    return _DZ_FIRST
+
+
 def _shape_seed(in_dims: int, out_dims: int) -> int:
     # Same spirit as reference layers.py:106-108 (seed from the layer
     # shape only): partition-invariant deterministic init.
@@ -228,7 +240,14 @@ class Linear(Module):
                                and B % 128 == 0 and B >= 128
                                and O >= 512 and I >= 512
                                and (O // 256) * (I // 256) >= 64)
-                if wide_glds or wgrad256_ok:
+                # dgrad-skipped layers (pipeline stage 0's first
+                # Linear): wgrad is the mask's only consumer and
+                # re-reads it once per N-tile — one elementwise dz
+                # pass is cheaper (A/B'd on the flagship first layer,
+                # SS_DZ_FIRST=0 reverts)
+                first_layer_dz = (not need_dx and B >= 4096
+                                  and _dz_first_enabled())
+                if wide_glds or wgrad256_ok or first_layer_dz:
                     dout = F.relu_bwd(dout, mask_src)
                     mask_src = None
         elif self.activation == "gelu":

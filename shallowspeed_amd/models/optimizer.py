@@ -39,6 +39,20 @@ def _split_tiled_transposes(params):
     return out
 
 
+def _build_blockmap(params, device, epb=16384):
+    """Block→(tensor, elem_base) map for the v2 fused optimizer
+    kernels: each block owns `epb` contiguous elements of one tensor,
+    replacing the v1 kernels' per-element binary search (a dependent
+    6-level desc walk that held the 523M-param wide-model step at
+    ~0.9 TB/s — profiles/r02_kernel_stats.md)."""
+    rows = []
+    for i, p in enumerate(params):
+        n = p.data.numel()
+        for b in range(0, n, epb):
+            rows.append([i, b])
+    return torch.tensor(rows, dtype=torch.int64).to(device)
+
+
 def _clip_flat(flat_grad, clip_norm):
     """Global-norm gradient clip on the stage's flat f32 grad buffer.
     Runs AFTER the DP all-reduce (grads are the summed global grads),
@@ -88,6 +102,7 @@ class SGD:
         cpu = torch.tensor(rows, dtype=torch.int64)
         self._desc = cpu.to(self.params[0].data.device)
         self._desc_cpu = cpu
+        self._bmap = _build_blockmap(self.params, self._desc.device)
 
     def step(self):
         if not self.params:
@@ -98,8 +113,8 @@ class SGD:
             ext = load_ext(required=True)
             if self._desc is None:
                 self._build_desc()
-            ext.sgd_multi(self._desc, self.lr, self._total, self.momentum,
-                          self.weight_decay)
+            ext.sgd_multi2(self._desc, self._bmap, self.lr, self.momentum,
+                           self.weight_decay)
             for _, lp, lpt in self._tiled_t:
                 ext.transpose_bf16(lp, lpt)
         else:
@@ -175,6 +190,7 @@ class AdamW:
         self._total = start
         cpu = torch.tensor(rows, dtype=torch.int64)
         self._desc = cpu.to(self.params[0].data.device)
+        self._bmap = _build_blockmap(self.params, self._desc.device)
 
     def step(self):
         if not self.params:
@@ -189,8 +205,8 @@ class AdamW:
             ext = load_ext(required=True)
             if self._desc is None:
                 self._build_desc()
-            ext.adamw_multi(self._desc, self.lr, self._total, b1, b2,
-                            self.eps, self.weight_decay, inv_bc1, inv_bc2)
+            ext.adamw_multi2(self._desc, self._bmap, self.lr, b1, b2,
+                             self.eps, self.weight_decay, inv_bc1, inv_bc2)
             for _, lp, lpt in self._tiled_t:
                 ext.transpose_bf16(lp, lpt)
         else:

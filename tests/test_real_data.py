@@ -34,3 +34,12 @@ def test_real_digits_convergence_cli(tmp_path):
     accs = [float(m) for m in re.findall(r"val_acc=([0-9.]+)", out)]
     assert accs, out
     assert max(accs) >= 0.95, f"best val_acc {max(accs)} < 0.95\n{out[-2000:]}"
+
+
+def test_lr_schedule_cli(tmp_path):
+    """Warmup + cosine schedule runs end to end through the CLI."""
+    out = _run([sys.executable, "train.py", "--epochs", "6", "--lr", "0.1",
+                "--lr-schedule", "cosine", "--warmup-epochs", "2",
+                "--loss", "xent", "--samples", "1024", "--device", "cpu"])
+    accs = [float(m) for m in re.findall(r"val_acc=([0-9.]+)", out)]
+    assert len(accs) >= 6 and accs[-1] > accs[0]

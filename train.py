@@ -79,6 +79,12 @@ def main():
                     help="global-norm gradient clipping threshold "
                          "(applied on the flat grad buffer after the DP "
                          "all-reduce, before the optimizer step)")
+    ap.add_argument("--lr-schedule", choices=["const", "cosine"],
+                    default="const",
+                    help="per-epoch LR schedule: const (reference "
+                         "behavior) or cosine decay to 10%% of --lr")
+    ap.add_argument("--warmup-epochs", type=int, default=0,
+                    help="linear LR warmup epochs before the schedule")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise run-to-run reproducible GPU training "
                          "(forces the single-owner split_k=1 wgrad path; "
@@ -180,8 +186,21 @@ def main():
            f"schedule={args.schedule} device={device} "
            f"layers={args.layer_sizes} loss={args.loss}")
 
+    def lr_at(epoch):
+        """Linear warmup then const/cosine decay (to 10% of peak)."""
+        if args.warmup_epochs and epoch < args.warmup_epochs:
+            return args.lr * (epoch + 1) / args.warmup_epochs
+        if args.lr_schedule == "cosine":
+            import math
+
+            span = max(1, args.epochs - args.warmup_epochs)
+            t = (epoch - args.warmup_epochs) / span
+            return args.lr * (0.1 + 0.9 * 0.5 * (1 + math.cos(math.pi * t)))
+        return args.lr
+
     for epoch in range(args.epochs):
         t0 = time.time()
+        optimizer.lr = lr_at(epoch)
         acc = compute_accuracy(model, val_worker, val_ds, topo)
         for batch_id in range(train_ds.num_batches()):
             sched = sched_cls(train_ds.num_mubatches(), args.pp, topo.stage_id)

@@ -44,7 +44,9 @@ def _dz_first_enabled() -> bool:
     if _DZ_FIRST is None:
         import os
 
-        _DZ_FIRST = os.environ.get("SS_DZ_FIRST", "1") == "1"
+        # default OFF: A/B'd neutral at the flagship (0.2072 vs 0.2081
+        # ms within one call) — the mask re-reads are L2-absorbed there
+        _DZ_FIRST = os.environ.get("SS_DZ_FIRST", "0") == "1"
     return _DZ_FIRST
 
 

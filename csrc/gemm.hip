@@ -661,7 +661,35 @@ __global__ __launch_bounds__(256) void colsum8_kernel(
     const int r1 = min(M, r0 + rows_per_block);
     if (col0 >= N) return;
     float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    for (int r = r0; r < r1; ++r) {
+    // 4-row unroll: four independent 16-B loads in flight per thread
+    // per iteration (the single-row loop was latency-serialized —
+    // measured 78 µs at 16384×4096 vs the ~17 µs traffic floor)
+    int r = r0;
+    for (; r + 4 <= r1; r += 4) {
+        bf16x8 v0 = *(const bf16x8*)(dY + (long)(r + 0) * N + col0);
+        bf16x8 v1 = *(const bf16x8*)(dY + (long)(r + 1) * N + col0);
+        bf16x8 v2 = *(const bf16x8*)(dY + (long)(r + 2) * N + col0);
+        bf16x8 v3 = *(const bf16x8*)(dY + (long)(r + 3) * N + col0);
+        if constexpr (HAS_MASK) {
+            bf16x8 m0 = *(const bf16x8*)(mask + (long)(r + 0) * N + col0);
+            bf16x8 m1 = *(const bf16x8*)(mask + (long)(r + 1) * N + col0);
+            bf16x8 m2 = *(const bf16x8*)(mask + (long)(r + 2) * N + col0);
+            bf16x8 m3 = *(const bf16x8*)(mask + (long)(r + 3) * N + col0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                if (bf2f(m0[e]) > 0.f) s[e] += bf2f(v0[e]);
+                if (bf2f(m1[e]) > 0.f) s[e] += bf2f(v1[e]);
+                if (bf2f(m2[e]) > 0.f) s[e] += bf2f(v2[e]);
+                if (bf2f(m3[e]) > 0.f) s[e] += bf2f(v3[e]);
+            }
+        } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                s[e] += bf2f(v0[e]) + bf2f(v1[e]) + bf2f(v2[e]) +
+                        bf2f(v3[e]);
+        }
+    }
+    for (; r < r1; ++r) {
         bf16x8 v = *(const bf16x8*)(dY + (long)r * N + col0);
         if constexpr (HAS_MASK) {
             bf16x8 m = *(const bf16x8*)(mask + (long)r * N + col0);

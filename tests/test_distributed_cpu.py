@@ -57,13 +57,18 @@ def _run_dist(fn, world, tmp_path, **kwargs):
 # ------------------------------------------------------------------ DP
 
 def _dp_train(rank, world, out_dir, loss="mse", schedule="naive",
-              bucket_bytes=25 << 20, force_defer=False):
+              bucket_bytes=25 << 20, force_defer=False, optimizer="sgd"):
     from shallowspeed_amd.parallel import init_topology
 
     topo = init_topology(dp=world, pp=1, backend="gloo",
                          device=torch.device("cpu"))
     model = MLP(SIZES, 0, 1, GBS, loss=loss).materialize_device("cpu")
-    opt = SGD(model.parameters(), lr=LR)
+    if optimizer == "adamw":
+        from shallowspeed_amd.models import AdamW
+
+        opt = AdamW(model.parameters(), lr=LR / 5, weight_decay=0.01)
+    else:
+        opt = SGD(model.parameters(), lr=LR)
     # local batch = GBS/world, split into MUB µbatches per rank
     ds = Dataset(GBS, (GBS // world) // MUB,
                  n_samples=N, in_dim=SIZES[0], n_classes=SIZES[-1])
@@ -117,6 +122,28 @@ def test_dp2_multibucket_matches_serial(tmp_path):
     want = _serial_params()
     for g, w in zip(got, want):
         torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)
+
+
+def test_dp2_adamw_matches_serial(tmp_path):
+    """AdamW under DP: identical summed grads on every replica must
+    produce identical moments and parameters (hash-sync holds for
+    stateful optimizers too), and match serial AdamW training."""
+    _run_dist(_dp_train, 2, tmp_path, schedule="gpipe", optimizer="adamw")
+    got = torch.load(tmp_path / "dp_params.pt", weights_only=False)
+
+    from shallowspeed_amd.models import AdamW
+    from shallowspeed_amd.parallel import NaiveParallelSchedule
+
+    model = MLP(SIZES, 0, 1, GBS, loss="mse").materialize_device("cpu")
+    opt = AdamW(model.parameters(), lr=LR / 5, weight_decay=0.01)
+    ds = Dataset(GBS, MUB, n_samples=N, in_dim=SIZES[0], n_classes=SIZES[-1])
+    ds.load(0, 1)
+    w = Worker(Topology(), model, ds, opt)
+    for b in range(ds.num_batches()):
+        w.execute(NaiveParallelSchedule(ds.num_mubatches(), 1, 0), b)
+    want = [p.data for p in model.parameters()]
+    for g, wv in zip(got, want):
+        torch.testing.assert_close(g, wv, rtol=1e-4, atol=1e-5)
 
 
 @pytest.mark.parametrize("schedule", ["naive", "gpipe", "pipedream"])

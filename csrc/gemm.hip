@@ -865,7 +865,16 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
         return e ? atoi(e) : -1;
     }();
     if (env_big >= 0) big = env_big != 0;
-    const int bm = big ? 128 : 64, bn = big ? 128 : 64;
+    // experimental 128x256 tile (halves dY re-reads when N spans many
+    // 128-tiles but Mo is narrow — the flagship first layer's
+    // 16384x256x784 masked wgrad): SS_WGRAD_N256=1 enables
+    static int env_n256 = []() {
+        const char* e = getenv("SS_WGRAD_N256");
+        return e ? atoi(e) : 0;
+    }();
+    const bool n256 = env_n256 && Mo >= 128 && N >= 640;
+    const int bm = n256 ? 128 : (big ? 128 : 64);
+    const int bn = n256 ? 256 : (big ? 128 : 64);
     // pick split for ~512 blocks total (2 per CU): more split
     // duplicates the atomic-epilogue traffic (measured: 1024² best at
     // exactly the coverage split; over-splitting by K depth cost
@@ -888,7 +897,9 @@ void ss_wgrad_tn(const void* dY, const void* X, const void* mask, void* gW,
                        stream, (const __bf16*)dY, (const __bf16*)X,         \
                        (const __bf16*)mask, (float*)gW, (float*)gb, Mo, N,  \
                        Kb, k_per_split, (const long*)nullptr, 1)
-    if (big) {
+    if (n256) {
+        if (mask) WLAUNCH(128, 256, true); else WLAUNCH(128, 256, false);
+    } else if (big) {
         if (mask) WLAUNCH(128, 128, true); else WLAUNCH(128, 128, false);
     } else {
         if (mask) WLAUNCH(64, 64, true); else WLAUNCH(64, 64, false);

@@ -74,3 +74,36 @@ def test_1f1b_stash_bound_any_shape(P, M):
         # µbatch completes its forward during warmup)
         assert peak == min(M, min(M, P - 1 - s) + 1)
         assert inflight == 0
+
+
+def test_p2p_coalescing_preserves_semantics():
+    """Worker._coalesce_p2p groups only CONSECUTIVE p2p instructions,
+    preserves their relative order inside the batch, and leaves every
+    non-p2p instruction in place — for every schedule/stage/µbatch
+    combination."""
+    from shallowspeed_amd.parallel.schedules import SCHEDULES
+    from shallowspeed_amd.parallel.worker import Worker, _BatchedP2P
+
+    coalesce = Worker._coalesce_p2p
+    P2P = Worker._P2P_TYPES
+
+    class W:  # minimal shim: _coalesce_p2p only touches _P2P_TYPES
+        _P2P_TYPES = P2P
+
+    for name, cls in SCHEDULES.items():
+        for stages in (2, 3, 4):
+            for mub in (1, 2, 4, 8):
+                for sid in range(stages):
+                    sched = cls(mub, stages, sid)
+                    for cmds in sched.steps():
+                        out = coalesce(W(), list(cmds))
+                        flat = []
+                        for c in out:
+                            if isinstance(c, _BatchedP2P):
+                                assert len(c.cmds) >= 2
+                                assert all(isinstance(x, P2P)
+                                           for x in c.cmds)
+                                flat += c.cmds
+                            else:
+                                flat.append(c)
+                        assert flat == list(cmds), (name, sid, cmds)

@@ -93,6 +93,11 @@ def main():
     ap.add_argument("--fp8", action="store_true",
                     help="MX-fp8 serving path for qualifying wide layers "
                          "(weights pre-quantized; activations per batch)")
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the forward-only batch in a hipGraph "
+                         "and replay it per batch (single GPU; collapses "
+                         "the per-kernel launch gaps that dominate the "
+                         "forward-only serving loop)")
     args = ap.parse_args()
 
     sizes = [int(s) for s in args.layer_sizes.split(",")]
@@ -122,6 +127,30 @@ def main():
         assert args.pp == 1, "--fp8 serving path is single-stage"
         fp8_plan = build_fp8_plan(model, args.batch)
 
+    graphed = None
+    if args.graph and device.type == "cuda" and topo.world == 1 \
+            and not fp8_plan:
+        # Serving graph: one persistent input slot + captured forward
+        # + argmax; per batch = one D2D copy + one graph replay.  The
+        # forward-only loop is launch-gap-bound (kernel sum ~59 µs vs
+        # ~91 µs wall at the flagship shape), so replay reclaims the
+        # gaps in a way the GPU-bound training step cannot.
+        from shallowspeed_amd.ops.functional import row_argmax as _ram
+
+        model.eval()
+        gx = torch.empty_like(ds.x_compute[:args.batch])
+        gx.copy_(ds.x_compute[:args.batch])
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            gout = _ram(model.forward(gx))
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            gout = _ram(model.forward(gx))
+        graphed = (g, gx, gout)
+
     preds = []
     # warmup
     if fp8_plan:
@@ -133,6 +162,13 @@ def main():
     from shallowspeed_amd.ops.functional import row_argmax
 
     for b in range(ds.num_batches()):
+        if graphed is not None:
+            g, gx, gout = graphed
+            gx.copy_(ds.x_compute[b * args.batch:(b + 1) * args.batch],
+                     non_blocking=True)
+            g.replay()
+            preds.append(gout.clone())
+            continue
         if fp8_plan:
             logits = fp8_forward(fp8_plan,
                                  ds.x[b * args.batch:(b + 1) * args.batch])

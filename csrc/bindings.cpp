@@ -44,7 +44,9 @@ void ss_gelu_fwd(const void*, void*, long, hipStream_t);
 void ss_gelu_bwd(const void*, const void*, void*, long, hipStream_t);
 void ss_row_argmax(const void*, void*, int, int, hipStream_t);
 }
-// C++-linkage (gemm256.hip / wgrad256.hip)
+// C++-linkage (gemm256.hip / wgrad256.hip / fused_mlp.hip)
+bool ss_fused_mlp_fwd(const void*, const void*, int, int, int, int, void*,
+                      hipStream_t);
 bool ss_gemm_nt_256w(const void*, const void*, const void*, void*, int, int,
                      int, bool, hipStream_t);
 bool ss_gemm_nt_256(const void*, const void*, const void*, void*, int, int,
@@ -456,6 +458,24 @@ void adamw_multi2(torch::Tensor desc, torch::Tensor bmap, double lr,
                     cur_stream());
 }
 
+torch::Tensor fused_mlp_argmax(torch::Tensor x, torch::Tensor desc,
+                               int64_t in_dim, int64_t nhidden,
+                               int64_t cout) {
+    check_bf16(x, "x");
+    TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kLong &&
+                    desc.is_contiguous() &&
+                    desc.numel() == (nhidden + 1) * 2,
+                "desc must be CUDA int64 [(nhidden+1)*2]");
+    const int M = x.size(0);
+    TORCH_CHECK(x.size(1) == in_dim, "in_dim mismatch");
+    auto out = torch::empty({M}, x.options().dtype(torch::kInt32));
+    TORCH_CHECK(ss_fused_mlp_fwd(x.data_ptr(), desc.data_ptr(), M,
+                                 (int)in_dim, (int)nhidden, (int)cout,
+                                 out.data_ptr(), cur_stream()),
+                "shape outside the fused-MLP tier");
+    return out;
+}
+
 void transpose_bf16(torch::Tensor src, torch::Tensor dst) {
     check_bf16(src, "src");
     check_bf16(dst, "dst");
@@ -499,6 +519,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("transpose_bf16", &transpose_bf16);
     m.def("sgd_multi2", &sgd_multi2);
     m.def("adamw_multi2", &adamw_multi2);
+    m.def("fused_mlp_argmax", &fused_mlp_argmax,
+          "persistent fused MLP forward + per-row argmax (serving)");
     m.def("ln_fwd", &ln_fwd);
     m.def("ln_bwd_dx", &ln_bwd_dx);
     m.def("ln_bwd_dparam", &ln_bwd_dparam);

@@ -52,6 +52,35 @@ def build_fp8_plan(model, batch):
     return plan
 
 
+def build_fused_plan(model, batch):
+    """Qualify the model for the persistent fused-MLP forward kernel
+    (csrc/fused_mlp.hip): 256-wide ReLU hidden chain (1..8 layers),
+    head ≤16 classes, batch % 64 == 0.  Activations stay LDS-resident
+    across the whole chain; the head emits per-row argmax directly."""
+    import torch
+
+    from shallowspeed_amd.models import Linear
+
+    ls = [l for l in model.layers if isinstance(l, Linear)]
+    if len(ls) < 2:
+        return None
+    hidden, head = ls[:-1], ls[-1]
+    ok = (batch % 64 == 0 and 1 <= len(hidden) <= 8
+          and all(l.out_dims == 256 and l.activation == "relu"
+                  for l in hidden)
+          and head.in_dims == 256 and head.out_dims <= 16
+          and head.activation is None
+          and model.device.type == "cuda")
+    if not ok:
+        return None
+    ptrs = []
+    for l in ls:
+        ptrs += [l.weight.compute().data_ptr(),
+                 l.bias.compute().data_ptr()]
+    desc = torch.tensor(ptrs, dtype=torch.int64).to(model.device)
+    return (desc, ls[0].in_dims, len(hidden), head.out_dims)
+
+
 def fp8_forward(plan, x):
     """fp8-RESIDENT chain: a layer whose successor is also fp8 emits
     its output pre-quantized from the GEMM epilogue (gemm_nt_f8_q) —
@@ -93,6 +122,11 @@ def main():
     ap.add_argument("--fp8", action="store_true",
                     help="MX-fp8 serving path for qualifying wide layers "
                          "(weights pre-quantized; activations per batch)")
+    ap.add_argument("--fused", action="store_true",
+                    help="persistent fused-MLP forward kernel (whole "
+                         "layer chain in ONE launch, LDS-resident "
+                         "activations, in-kernel argmax; 256-wide "
+                         "hidden chains)")
     ap.add_argument("--graph", action="store_true",
                     help="capture the forward-only batch in a hipGraph "
                          "and replay it per batch (single GPU; collapses "
@@ -127,9 +161,16 @@ def main():
         assert args.pp == 1, "--fp8 serving path is single-stage"
         fp8_plan = build_fp8_plan(model, args.batch)
 
+    fused_plan = None
+    if args.fused and topo.world == 1 and not fp8_plan:
+        fused_plan = build_fused_plan(model, args.batch)
+        if fused_plan is None:
+            print("# --fused: model/batch outside the fused tier, "
+                  "falling back", flush=True)
+
     graphed = None
     if args.graph and device.type == "cuda" and topo.world == 1 \
-            and not fp8_plan:
+            and not fp8_plan and fused_plan is None:
         # Serving graph: one persistent input slot + captured forward
         # + argmax; per batch = one D2D copy + one graph replay.  The
         # forward-only loop is launch-gap-bound (kernel sum ~59 µs vs
@@ -161,7 +202,17 @@ def main():
     t0 = time.perf_counter()
     from shallowspeed_amd.ops.functional import row_argmax
 
+    if fused_plan is not None:
+        from shallowspeed_amd.ops import load_ext
+
+        _e = load_ext(required=True)
+
     for b in range(ds.num_batches()):
+        if fused_plan is not None:
+            desc, ind, nh, co = fused_plan
+            xb = ds.x_compute[b * args.batch:(b + 1) * args.batch]
+            preds.append(_e.fused_mlp_argmax(xb, desc, ind, nh, co))
+            continue
         if graphed is not None:
             g, gx, gout = graphed
             gx.copy_(ds.x_compute[b * args.batch:(b + 1) * args.batch],

@@ -80,6 +80,7 @@ def build_ext(verbose=True):
         os.path.join(src_dir, "fp8.hip"),
         os.path.join(src_dir, "elementwise.hip"),
         os.path.join(src_dir, "norm.hip"),
+        os.path.join(src_dir, "fused_mlp.hip"),
     ]
     mod = load(
         name=_EXT_NAME,

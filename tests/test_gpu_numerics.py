@@ -695,3 +695,33 @@ def test_fp8_fused_quant_output(gpu_device):
     cd = sgn * mag * torch.pow(2.0, sc - 127)
     rel = ((cd - c16).abs() / c16.abs().clamp_min(4.0)).max().item()
     assert rel < 0.07, rel
+
+
+def test_fused_mlp_fwd_matches_eager(gpu_device):
+    """Persistent fused-MLP forward (csrc/fused_mlp.hip): same bf16
+    rounding points and k-accumulation order as the eager layer chain,
+    so per-row argmax must match exactly."""
+    from shallowspeed_amd.models import MLP
+    from shallowspeed_amd.ops import load_ext
+    from shallowspeed_amd.ops.functional import row_argmax
+
+    e = load_ext(required=True)
+    for sizes, rows in (([784, 256, 256, 256, 10], 1024),
+                        ([784, 256, 256, 256, 256, 256, 256, 10], 448),
+                        ([100, 256, 10], 64)):
+        model = MLP(sizes, 0, 1, rows, loss="xent").materialize_device(
+            gpu_device)
+        model.eval()
+        g = torch.Generator().manual_seed(len(sizes))
+        x = torch.randn(rows, sizes[0], generator=g).bfloat16().to(gpu_device)
+        probs = model.forward(x)
+        want = row_argmax(probs)
+        import infer as infer_mod
+
+        plan = infer_mod.build_fused_plan(model, rows)
+        assert plan is not None, sizes
+        desc, ind, nh, co = plan
+        got = e.fused_mlp_argmax(x, desc, ind, nh, co)
+        torch.cuda.synchronize()
+        mismatch = (got.long() != want.long()).sum().item()
+        assert mismatch == 0, (sizes, mismatch)

@@ -723,5 +723,19 @@ def test_fused_mlp_fwd_matches_eager(gpu_device):
         desc, ind, nh, co = plan
         got = e.fused_mlp_argmax(x, desc, ind, nh, co)
         torch.cuda.synchronize()
-        mismatch = (got.long() != want.long()).sum().item()
-        assert mismatch == 0, (sizes, mismatch)
+        # The fused head argmaxes UNROUNDED f32 logits; the eager path
+        # argmaxes bf16-rounded probs.  On a random-init model the
+        # near-uniform probs produce exact bf16 ties, where the two may
+        # legally pick different (equal-prob) classes — every
+        # disagreement must be such a tie; anything else is a bug.
+        gl = got.long()
+        wl = want.long()
+        mm = (gl != wl).nonzero().flatten()
+        rows_idx = torch.arange(probs.shape[0], device=probs.device)
+        p_got = probs[rows_idx, gl]
+        p_want = probs[rows_idx, wl]
+        if mm.numel():
+            assert torch.equal(p_got[mm], p_want[mm]), (
+                sizes, mm[:5].tolist())
+        assert mm.numel() <= max(1, probs.shape[0] // 50), (
+            sizes, mm.numel())  # ties must stay rare

@@ -120,51 +120,28 @@ __global__ __launch_bounds__(256, 1) void fused_mlp_fwd_kernel(
                 if (l == 0) ra.load(X, M, K, m0, (t + 1) * KS, tid);
                 rb.load(W, brows, K, 0, (t + 1) * KS, tid);
             }
-            if (l < nhidden) {
 #pragma unroll
-                for (int kk = 0; kk < KS / 32; ++kk) {
-                    bf16x8 a_frag[4], b_frag[4];
+            for (int kk = 0; kk < KS / 32; ++kk) {
+                bf16x8 a_frag[4], b_frag[4];
 #pragma unroll
-                    for (int i = 0; i < 4; ++i) {
-                        if (l == 0)
-                            a_frag[i] = *(const bf16x8*)&axsp
-                                [cur * 64 + i * 16 + lrow][kk * 32 + kch * 8];
-                        else
-                            a_frag[i] = *(const bf16x8*)&act[cur_act]
-                                [i * 16 + lrow][t * KS + kk * 32 + kch * 8];
-                    }
-#pragma unroll
-                    for (int j = 0; j < 4; ++j)
-                        b_frag[j] = *(const bf16x8*)&ws[cur]
-                            [wave * 64 + j * 16 + lrow][kk * 32 + kch * 8];
-#pragma unroll
-                    for (int i = 0; i < 4; ++i)
-#pragma unroll
-                        for (int j = 0; j < 4; ++j)
-                            acc[i][j] =
-                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    a_frag[i], b_frag[j], acc[i][j], 0, 0,
-                                    0);
-                }
-            } else if (wave == 0) {
-                // head: only the first 16 output columns are real
-                // (cout <= 16) — waves 1-3 idle through the (tiny)
-                // head GEMM instead of computing unused columns
-                // (~20% of total MFMA instructions, PMC-measured)
-#pragma unroll
-                for (int kk = 0; kk < KS / 32; ++kk) {
-                    bf16x8 a_frag[4], b_frag;
-#pragma unroll
-                    for (int i = 0; i < 4; ++i)
+                for (int i = 0; i < 4; ++i) {
+                    if (l == 0)
+                        a_frag[i] = *(const bf16x8*)&axsp
+                            [cur * 64 + i * 16 + lrow][kk * 32 + kch * 8];
+                    else
                         a_frag[i] = *(const bf16x8*)&act[cur_act]
                             [i * 16 + lrow][t * KS + kk * 32 + kch * 8];
-                    b_frag = *(const bf16x8*)&ws[cur][lrow]
-                                                     [kk * 32 + kch * 8];
-#pragma unroll
-                    for (int i = 0; i < 4; ++i)
-                        acc[i][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            a_frag[i], b_frag, acc[i][0], 0, 0, 0);
                 }
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    b_frag[j] = *(const bf16x8*)&ws[cur]
+                        [wave * 64 + j * 16 + lrow][kk * 32 + kch * 8];
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+#pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
             }
             if (t + 1 < nsteps) {
                 if (l == 0) ra.write(axsp + (cur ^ 1) * 64, tid);

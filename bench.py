@@ -162,6 +162,8 @@ def main():
                 "mubatches": args.mubatches,
                 "hipgraph": use_graph,
                 "fp8_fwd": args.fp8_fwd,
+                "peak_mem_gb": (round(torch.cuda.max_memory_allocated()
+                                      / 2**30, 2) if is_gpu else None),
             },
         }), flush=True)
 

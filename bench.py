@@ -153,7 +153,11 @@ def main():
             "dtype": "bf16" if is_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "mlp4_mnist_784x256x256x256x10",
+                "model": ("mlp4_mnist_784x256x256x256x10"
+                          if sizes == [784, 256, 256, 256, 10] else
+                          "mlp%d_%s" % (len(sizes) - 1,
+                                        "x".join(map(str, sizes[:3]))
+                                        + ("..." if len(sizes) > 4 else ""))),
                 "global_batch": global_batch,
                 "seq_len": None,
                 "parallelism": f"dp{dp}" + (f"xpp{pp}" if pp > 1 else ""),

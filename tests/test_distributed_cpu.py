@@ -24,10 +24,11 @@ LR = 0.05
 STEPS_BATCHES = 2
 
 
-def _serial_params(loss="mse", schedule="naive"):
-    model = MLP(SIZES, 0, 1, GBS, loss=loss).materialize_device("cpu")
+def _serial_params(loss="mse", schedule="naive", sizes=None):
+    sizes = sizes or SIZES
+    model = MLP(sizes, 0, 1, GBS, loss=loss).materialize_device("cpu")
     opt = SGD(model.parameters(), lr=LR)
-    ds = Dataset(GBS, MUB, n_samples=N, in_dim=SIZES[0], n_classes=SIZES[-1])
+    ds = Dataset(GBS, MUB, n_samples=N, in_dim=sizes[0], n_classes=sizes[-1])
     ds.load(0, 1)
     w = Worker(Topology(), model, ds, opt)
     for b in range(ds.num_batches()):
@@ -161,16 +162,17 @@ def test_dp2_deferred_wgrad_matches_serial(tmp_path, schedule):
 
 # ------------------------------------------------------------------ PP
 
-def _pp_train(rank, world, out_dir, schedule="gpipe"):
+def _pp_train(rank, world, out_dir, schedule="gpipe", sizes=None):
     from shallowspeed_amd.parallel import init_topology
 
+    sizes = sizes or SIZES
     topo = init_topology(dp=1, pp=world, backend="gloo",
                          device=torch.device("cpu"))
-    model = MLP(SIZES, topo.stage_id, world, GBS,
+    model = MLP(sizes, topo.stage_id, world, GBS,
                 loss="mse").materialize_device("cpu")
     opt = SGD(model.parameters(), lr=LR)
-    ds = Dataset(GBS, GBS // MUB, n_samples=N, in_dim=SIZES[0],
-                 n_classes=SIZES[-1])
+    ds = Dataset(GBS, GBS // MUB, n_samples=N, in_dim=sizes[0],
+                 n_classes=sizes[-1])
     ds.load(0, 1)
     w = Worker(topo, model, ds, opt)
     cls = SCHEDULES[schedule]
@@ -234,6 +236,20 @@ def test_dp2xpp2_matches_serial(tmp_path):
     for s in range(2):
         got += torch.load(tmp_path / f"grid_stage{s}.pt", weights_only=False)
     want = _serial_params()
+    assert len(got) == len(want)
+    for g, w in zip(got, want):
+        torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)
+
+
+def test_pp4_1f1b_matches_serial(tmp_path):
+    """Deepest CPU-testable pipeline: 4 stages, 1F1B, world 4
+    (8 layer boundaries so len(sizes) divides into 4 stages)."""
+    sizes8 = [24, 16, 12, 8, 6, 12, 8, 10]
+    _run_dist(_pp_train, 4, tmp_path, schedule="pipedream", sizes=sizes8)
+    got = []
+    for s in range(4):
+        got += torch.load(tmp_path / f"pp_stage{s}.pt", weights_only=False)
+    want = _serial_params(sizes=sizes8)
     assert len(got) == len(want)
     for g, w in zip(got, want):
         torch.testing.assert_close(g, w, rtol=1e-4, atol=1e-5)

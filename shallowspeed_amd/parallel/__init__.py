@@ -9,3 +9,9 @@ from .schedules import (  # noqa: F401
     Schedule,
 )
 from .worker import Worker  # noqa: F401
+from .tp import (  # noqa: F401
+    TPMLP,
+    ColumnParallelLinear,
+    RowParallelLinear,
+    tp_mlp_layers,
+)

@@ -39,8 +39,10 @@ class Topology:
     world: int = 1
     dp: int = 1
     pp: int = 1
+    tp: int = 1
     dp_group: Optional[object] = None
     pp_group: Optional[object] = None
+    tp_group: Optional[object] = None
     device: torch.device = field(default_factory=lambda: torch.device("cpu"))
 
     @property
@@ -53,7 +55,10 @@ class Topology:
 
     @property
     def dp_rank(self):
-        return self.pipeline_id
+        # modulo dp so pure-TP topologies (dp=1, world=tp) read the
+        # SAME data shard on every rank (the batch is replicated
+        # across TP; rank // pp alone would shift rank 1's shard)
+        return (self.rank // self.pp) % self.dp
 
     @property
     def prev_rank(self):
@@ -70,15 +75,22 @@ class Topology:
 
 
 def init_topology(dp: int, pp: int, backend: Optional[str] = None,
-                  device: Optional[torch.device] = None) -> Topology:
-    """Initialize the process group and the DP/PP subgroup grid.
+                  device: Optional[torch.device] = None,
+                  tp: int = 1) -> Topology:
+    """Initialize the process group and the DP/PP subgroup grid (or,
+    with tp>1, a pure tensor-parallel group over the whole world —
+    TP composes with dp=pp=1 in this round).
 
     Single-process (dp=pp=1, no env rendezvous) returns a trivial
     topology without touching torch.distributed.
     """
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
-    assert dp * pp == world, f"DP({dp})×PP({pp}) != world({world})"
+    if tp > 1:
+        assert dp == 1 and pp == 1, "TP currently composes with dp=pp=1"
+        assert tp == world, f"TP({tp}) != world({world})"
+    else:
+        assert dp * pp == world, f"DP({dp})×PP({pp}) != world({world})"
 
     if device is None:
         if torch.cuda.is_available():
@@ -111,6 +123,16 @@ def init_topology(dp: int, pp: int, backend: Optional[str] = None,
             # where cross-rank first-use ordering can deadlock.
             kw["device_id"] = device
         dist.init_process_group(**kw)
+
+    if tp > 1:
+        topo = Topology(rank=rank, world=world, dp=1, pp=1, tp=tp,
+                        tp_group=dist.group.WORLD, device=device)
+        _barrier(topo)
+        t = torch.ones(1, device=device if device.type == "cuda"
+                       else torch.device("cpu"))
+        dist.all_reduce(t, group=topo.tp_group)  # eager comm creation
+        _barrier(topo)
+        return topo
 
     # orthogonal subgroups; EVERY rank must call new_group for EVERY
     # group in the same order.

@@ -1,0 +1,77 @@
+"""Tensor parallelism (beyond-reference): column/row-parallel Linear
+pair equivalence vs the serial model — single-process (tp_world=1
+degenerates to serial math) and multi-process (gloo, world 2/4)."""
+
+import os
+import random
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from shallowspeed_amd.data import Dataset
+from shallowspeed_amd.models import MLP, SGD
+from shallowspeed_amd.parallel import TPMLP, NaiveParallelSchedule, Topology, Worker
+
+SIZES = [24, 32, 16, 10]  # col(24->32) row(32->16) + replicated head
+GBS, N, LR = 16, 64, 0.05
+
+
+def _serial(loss="xent", steps=3):
+    model = MLP(SIZES, 0, 1, GBS, loss=loss).materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=LR)
+    ds = Dataset(GBS, GBS, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1]).load(0, 1)
+    w = Worker(Topology(), model, ds, opt)
+    for b in range(min(steps, ds.num_batches())):
+        w.execute(NaiveParallelSchedule(1, 1, 0), b)
+    model.eval()
+    g = torch.Generator().manual_seed(42)
+    xin = torch.randn(8, SIZES[0], generator=g)
+    return model.forward(xin), xin
+
+
+def test_tp1_equals_serial_mlp():
+    """tp_world=1: the TP model is numerically the serial model."""
+    want, xin = _serial(steps=0)
+    tp = TPMLP(SIZES, None, 0, 1, GBS).materialize_device("cpu")
+    tp.eval()
+    torch.testing.assert_close(tp.forward(xin), want,
+                               rtol=1e-5, atol=1e-6)
+
+
+def _tp_entry(rank, world, port, out_dir):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo")
+    group = dist.group.WORLD
+    model = TPMLP(SIZES, group, rank, world, GBS).materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=LR)
+    ds = Dataset(GBS, GBS, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1]).load(0, 1)
+    w = Worker(Topology(rank=rank, world=world), model, ds, opt,
+               use_dp=False)
+    for b in range(3):
+        w.execute(NaiveParallelSchedule(1, 1, 0), b)
+    model.eval()
+    g = torch.Generator().manual_seed(42)
+    xin = torch.randn(8, SIZES[0], generator=g)
+    out = model.forward(xin)
+    if rank == 0:
+        torch.save(out, os.path.join(out_dir, "tp_out.pt"))
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_tp_training_matches_serial(tmp_path, world):
+    """TP=2/4 training (3 steps) produces the same function as serial
+    training: forward on a fixed probe input matches within f32
+    reduction-order tolerance."""
+    port = random.randint(20000, 45000)
+    mp.spawn(_tp_entry, args=(world, port, str(tmp_path)), nprocs=world,
+             join=True)
+    got = torch.load(tmp_path / "tp_out.pt", weights_only=False)
+    want, _ = _serial(steps=3)
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)

@@ -65,6 +65,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--dp", type=int, default=1)
     ap.add_argument("--pp", type=int, default=1)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor parallelism (Megatron-style column/row "
+                         "pair, beyond-reference); composes with dp=pp=1")
     ap.add_argument("--schedule", choices=list(SCHEDULES), default="gpipe")
     ap.add_argument("--epochs", type=int, default=20)
     ap.add_argument("--global-batch", type=int, default=128)
@@ -121,7 +124,7 @@ def main():
         "len(--layer-sizes) must divide into --pp stages (layers.py:242)"
     device = torch.device(args.device) if args.device else None
     topo = init_topology(args.dp, args.pp, backend=args.backend,
-                         device=device)
+                         device=device, tp=args.tp)
     device = topo.device
 
     # model: stage slice of the full MLP (reference train.py:99-107)
@@ -130,8 +133,15 @@ def main():
 
         _F.set_deterministic(True)
 
-    model = MLP(args.layer_sizes, stage_idx=topo.stage_id, n_stages=args.pp,
-                global_batch_size=args.global_batch, loss=args.loss)
+    if args.tp > 1:
+        from shallowspeed_amd.parallel import TPMLP
+
+        model = TPMLP(args.layer_sizes, topo.tp_group, topo.rank, args.tp,
+                      args.global_batch, loss=args.loss)
+    else:
+        model = MLP(args.layer_sizes, stage_idx=topo.stage_id,
+                    n_stages=args.pp,
+                    global_batch_size=args.global_batch, loss=args.loss)
     model.materialize_device(device)
     clip_kw = dict(clip_norm=args.clip_grad, flat_grad=model._flat_grad)
     if args.optimizer == "adamw":
@@ -207,12 +217,14 @@ def main():
             worker.execute(sched, batch_id)
         if device.type == "cuda":
             torch.cuda.synchronize(device)
-        if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0:
+        if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0 \
+                and (topo.tp == 1 or topo.rank == 0):
             print(f"epoch {epoch:3d}  val_acc={acc:.4f}  "
                   f"time={time.time()-t0:.2f}s", flush=True)
 
     acc = compute_accuracy(model, val_worker, val_ds, topo)
-    if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0:
+    if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0 \
+            and (topo.tp == 1 or topo.rank == 0):
         print(f"final val_acc={acc:.4f}", flush=True)
 
     if args.save:
@@ -228,7 +240,8 @@ def main():
                            key=lambda kv: -kv[1]):
             print(f"  {k:24s} {v:8.3f}s ({v/total*100:5.1f}%)")
 
-    # replica-sync invariant (reference train.py:154-155)
+    # replica-sync invariant (reference train.py:154-155).  TP ranks
+    # hold different shards by design — no hash sync there.
     if topo.dp > 1:
         assert_sync(topo.dp_group, get_model_hash(model))
         rprint("DP replicas in sync ✓")

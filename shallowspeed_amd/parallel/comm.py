@@ -47,18 +47,25 @@ class Topology:
 
     @property
     def stage_id(self):
-        return self.rank % self.pp
+        return (self.rank // self.tp) % self.pp
 
     @property
     def pipeline_id(self):
         return self.rank // self.pp
 
     @property
+    def tp_rank(self):
+        # TP is the INNERMOST axis: consecutive ranks form one TP
+        # group (xGMI-adjacent on a standard node layout), DP strides
+        # across groups
+        return self.rank % self.tp
+
+    @property
     def dp_rank(self):
-        # modulo dp so pure-TP topologies (dp=1, world=tp) read the
-        # SAME data shard on every rank (the batch is replicated
-        # across TP; rank // pp alone would shift rank 1's shard)
-        return (self.rank // self.pp) % self.dp
+        # modulo dp so TP topologies read the SAME data shard on every
+        # rank of a TP group (the batch is replicated across TP); with
+        # tp=1 this is the original PP-major rank // pp
+        return (self.rank // (self.pp * self.tp)) % self.dp
 
     @property
     def prev_rank(self):
@@ -87,8 +94,8 @@ def init_topology(dp: int, pp: int, backend: Optional[str] = None,
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     if tp > 1:
-        assert dp == 1 and pp == 1, "TP currently composes with dp=pp=1"
-        assert tp == world, f"TP({tp}) != world({world})"
+        assert pp == 1, "TP composes with DP (pp=1) this round"
+        assert dp * tp == world, f"DP({dp})×TP({tp}) != world({world})"
     else:
         assert dp * pp == world, f"DP({dp})×PP({pp}) != world({world})"
 
@@ -125,12 +132,27 @@ def init_topology(dp: int, pp: int, backend: Optional[str] = None,
         dist.init_process_group(**kw)
 
     if tp > 1:
-        topo = Topology(rank=rank, world=world, dp=1, pp=1, tp=tp,
-                        tp_group=dist.group.WORLD, device=device)
+        # DP×TP grid, TP innermost: tp group d = [d*tp .. d*tp+tp),
+        # dp group t = [t, t+tp, t+2tp, ...).  EVERY rank calls
+        # new_group for EVERY group in the same order.
+        tp_groups = {}
+        for d in range(dp):
+            tp_groups[d] = dist.new_group([d * tp + t for t in range(tp)])
+        dp_groups = {}
+        for t in range(tp):
+            dp_groups[t] = dist.new_group([d * tp + t for d in range(dp)])
+        topo = Topology(rank=rank, world=world, dp=dp, pp=1, tp=tp,
+                        dp_group=dp_groups[rank % tp] if dp > 1 else None,
+                        tp_group=tp_groups[rank // tp], device=device)
+        # deterministic-order eager communicator creation (disjoint
+        # groups per phase, world barriers between phases)
+        dev = device if device.type == "cuda" else torch.device("cpu")
+        t1 = torch.ones(1, device=dev)
         _barrier(topo)
-        t = torch.ones(1, device=device if device.type == "cuda"
-                       else torch.device("cpu"))
-        dist.all_reduce(t, group=topo.tp_group)  # eager comm creation
+        if dp > 1:
+            dist.all_reduce(t1, group=topo.dp_group)
+        _barrier(topo)
+        dist.all_reduce(t1, group=topo.tp_group)
         _barrier(topo)
         return topo
 

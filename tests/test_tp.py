@@ -126,3 +126,43 @@ def test_tp2_gpu_matches_serial(tmp_path, gpu_device):
     want = model.forward(xin).float().cpu()
     torch.cuda.synchronize()
     torch.testing.assert_close(got, want, atol=5e-3, rtol=5e-2)
+
+
+def _dptp_entry(rank, world, port, out_dir, dp, tp):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    from shallowspeed_amd.parallel import init_topology
+    from shallowspeed_amd.utils import assert_sync, get_model_hash
+
+    topo = init_topology(dp=dp, pp=1, backend="gloo",
+                         device=torch.device("cpu"), tp=tp)
+    model = TPMLP(SIZES, topo.tp_group, topo.tp_rank, tp,
+                  GBS).materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=LR)
+    ds = Dataset(GBS, GBS // dp, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1]).load(topo.dp_rank, dp)
+    w = Worker(topo, model, ds, opt)
+    for b in range(ds.num_batches()):
+        w.execute(NaiveParallelSchedule(1, 1, 0), b)
+    # DP replicas of the SAME shard must be hash-identical
+    assert_sync(topo.dp_group, get_model_hash(model))
+    model.eval()
+    g = torch.Generator().manual_seed(42)
+    xin = torch.randn(8, SIZES[0], generator=g)
+    out = model.forward(xin)
+    if rank == 0:
+        torch.save(out, os.path.join(out_dir, "dptp_out.pt"))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+def test_dp2_tp2_matches_serial(tmp_path):
+    """The composed DP×TP grid (world 4, TP innermost): per-shard DP
+    hash sync + the trained function matches serial training."""
+    port = random.randint(20000, 45000)
+    mp.spawn(_dptp_entry, args=(4, port, str(tmp_path), 2, 2), nprocs=4,
+             join=True)
+    got = torch.load(tmp_path / "dptp_out.pt", weights_only=False)
+    want, _ = _serial(steps=4)
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)

@@ -67,7 +67,8 @@ def main():
     ap.add_argument("--pp", type=int, default=1)
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor parallelism (Megatron-style column/row "
-                         "pair, beyond-reference); composes with dp=pp=1")
+                         "pair, beyond-reference); composes with --dp "
+                         "(TP innermost = xGMI-adjacent ranks)")
     ap.add_argument("--schedule", choices=list(SCHEDULES), default="gpipe")
     ap.add_argument("--epochs", type=int, default=20)
     ap.add_argument("--global-batch", type=int, default=128)
@@ -136,8 +137,8 @@ def main():
     if args.tp > 1:
         from shallowspeed_amd.parallel import TPMLP
 
-        model = TPMLP(args.layer_sizes, topo.tp_group, topo.rank, args.tp,
-                      args.global_batch, loss=args.loss)
+        model = TPMLP(args.layer_sizes, topo.tp_group, topo.tp_rank,
+                      args.tp, args.global_batch, loss=args.loss)
     else:
         model = MLP(args.layer_sizes, stage_idx=topo.stage_id,
                     n_stages=args.pp,
@@ -218,13 +219,13 @@ def main():
         if device.type == "cuda":
             torch.cuda.synchronize(device)
         if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0 \
-                and (topo.tp == 1 or topo.rank == 0):
+                and (topo.tp == 1 or topo.tp_rank == 0):
             print(f"epoch {epoch:3d}  val_acc={acc:.4f}  "
                   f"time={time.time()-t0:.2f}s", flush=True)
 
     acc = compute_accuracy(model, val_worker, val_ds, topo)
     if topo.stage_id == topo.pp - 1 and topo.dp_rank == 0 \
-            and (topo.tp == 1 or topo.rank == 0):
+            and (topo.tp == 1 or topo.tp_rank == 0):
         print(f"final val_acc={acc:.4f}", flush=True)
 
     if args.save:

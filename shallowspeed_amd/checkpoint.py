@@ -24,6 +24,15 @@ import os
 import torch
 
 
+def _shard_name(topo):
+    """stage_{s}.pt, plus a _tp{t} suffix when tensor-parallel shards
+    exist (every TP rank owns different weights)."""
+    name = f"stage_{topo.stage_id:02d}"
+    if getattr(topo, "tp", 1) > 1:
+        name += f"_tp{topo.tp_rank:02d}"
+    return name + ".pt"
+
+
 def save_checkpoint(path, model, topo, step: int = 0, extra=None,
                     optimizer=None):
     os.makedirs(path, exist_ok=True)
@@ -38,9 +47,10 @@ def save_checkpoint(path, model, topo, step: int = 0, extra=None,
             # stateless SGD (no momentum): nothing worth persisting
             if any(k in opt_state for k in ("velocity", "exp_avg")):
                 state["opt"] = opt_state
-        torch.save(state, os.path.join(path, f"stage_{topo.stage_id:02d}.pt"))
+        torch.save(state, os.path.join(path, _shard_name(topo)))
     if topo.rank == 0:
-        meta = {"dp": topo.dp, "pp": topo.pp, "step": step}
+        meta = {"dp": topo.dp, "pp": topo.pp,
+                "tp": getattr(topo, "tp", 1), "step": step}
         if extra:
             meta.update(extra)
         torch.save(meta, os.path.join(path, "meta.pt"))
@@ -55,7 +65,12 @@ def load_checkpoint(path, model, topo, optimizer=None):
             f"checkpoint was written with pp={meta['pp']}; current "
             f"topology has pp={topo.pp} — stage shards do not repartition"
         )
-    f = os.path.join(path, f"stage_{topo.stage_id:02d}.pt")
+    if meta.get("tp", 1) != getattr(topo, "tp", 1):
+        raise AssertionError(
+            f"checkpoint was written with tp={meta.get('tp', 1)}; current "
+            f"topology has tp={getattr(topo, 'tp', 1)} — TP shards do "
+            f"not repartition")
+    f = os.path.join(path, _shard_name(topo))
     state = torch.load(f, map_location="cpu", weights_only=False)
     params = model.parameters()
     assert len(params) == len(state["params"]), (

@@ -166,3 +166,48 @@ def test_dp2_tp2_matches_serial(tmp_path):
     got = torch.load(tmp_path / "dptp_out.pt", weights_only=False)
     want, _ = _serial(steps=4)
     torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)
+
+
+def _tp_ckpt_entry(rank, world, port, out_dir):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    from shallowspeed_amd.checkpoint import load_checkpoint, save_checkpoint
+    from shallowspeed_amd.parallel import init_topology
+
+    topo = init_topology(dp=1, pp=1, backend="gloo",
+                         device=torch.device("cpu"), tp=world)
+    model = TPMLP(SIZES, topo.tp_group, topo.tp_rank, world,
+                  GBS).materialize_device("cpu")
+    opt = SGD(model.parameters(), lr=LR)
+    ds = Dataset(GBS, GBS, n_samples=N, in_dim=SIZES[0],
+                 n_classes=SIZES[-1]).load(0, 1)
+    w = Worker(topo, model, ds, opt, use_dp=False)
+    for b in range(2):
+        w.execute(NaiveParallelSchedule(1, 1, 0), b)
+    save_checkpoint(os.path.join(out_dir, "ck"), model, topo, step=2)
+
+    fresh = TPMLP(SIZES, topo.tp_group, topo.tp_rank, world,
+                  GBS).materialize_device("cpu")
+    load_checkpoint(os.path.join(out_dir, "ck"), fresh, topo)
+    for a, b2 in zip(model.parameters(), fresh.parameters()):
+        torch.testing.assert_close(a.data, b2.data, rtol=0, atol=0)
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+def test_tp_checkpoint_shards_per_rank(tmp_path):
+    """TP checkpoints write one shard file PER TP RANK (no stage_00
+    collision) and round-trip exactly; tp mismatch on load is
+    rejected."""
+    port = random.randint(20000, 45000)
+    mp.spawn(_tp_ckpt_entry, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    files = sorted(os.listdir(tmp_path / "ck"))
+    assert "stage_00_tp00.pt" in files and "stage_00_tp01.pt" in files
+    # tp-mismatch rejection (serial topology reading a tp=2 checkpoint)
+    from shallowspeed_amd.checkpoint import load_checkpoint
+
+    m = TPMLP(SIZES, None, 0, 1, GBS).materialize_device("cpu")
+    with pytest.raises(AssertionError, match="tp="):
+        load_checkpoint(str(tmp_path / "ck"), m, Topology())

@@ -14,9 +14,13 @@ assert_sync).  Layout:
 
 Resuming a momentum>0 run restores the exact optimization trajectory
 (velocities + hyperparameters are saved with the stage shard); loading
-validates the checkpoint's PP topology against the current one so a
+validates the checkpoint's PP/TP topology against the current one so a
 repartitioned resume fails loudly instead of via per-tensor shape
-asserts alone.
+asserts alone.  TP shards write one file per TP rank
+(stage_XX_tpYY.pt).  If ranks load a checkpoint IMMEDIATELY after a
+collective save in the same run, put a dist.barrier() between save and
+load — rank 0 writes meta.pt and other ranks must not race it (the
+normal save-at-exit / resume-at-start flow never hits this).
 """
 
 import os

@@ -185,6 +185,9 @@ def _tp_ckpt_entry(rank, world, port, out_dir):
     for b in range(2):
         w.execute(NaiveParallelSchedule(1, 1, 0), b)
     save_checkpoint(os.path.join(out_dir, "ck"), model, topo, step=2)
+    import torch.distributed as dist
+
+    dist.barrier()  # loaders must not race rank 0's meta.pt write
 
     fresh = TPMLP(SIZES, topo.tp_group, topo.tp_rank, world,
                   GBS).materialize_device("cpu")

@@ -27,6 +27,9 @@ def main():
     ap.add_argument("--local-batch", type=int, default=16384)
     ap.add_argument("--mubatches", type=int, default=1)
     ap.add_argument("--pp", type=int, default=1)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor parallelism (composes with --dp; "
+                         "pp must be 1)")
     ap.add_argument("--schedule", default="naive",
                     choices=["naive", "gpipe", "pipedream"])
     ap.add_argument("--layer-sizes", default="784,256,256,256,10")
@@ -61,11 +64,18 @@ def main():
         f"--gpus {args.gpus} vs WORLD_SIZE {world}"
     n = max(world, 1)
     pp = args.pp
-    dp = n // pp
-    assert dp * pp == n, f"dp({dp})*pp({pp}) != {n}"
+    tp = args.tp
+    if tp > 1:
+        assert pp == 1, "tp composes with dp only"
+        dp = n // tp
+        assert dp * tp == n, f"dp({dp})*tp({tp}) != {n}"
+    else:
+        dp = n // pp
+        assert dp * pp == n, f"dp({dp})*pp({pp}) != {n}"
 
     device = torch.device(args.device) if args.device else None
-    topo = init_topology(dp, pp, backend=args.backend, device=device)
+    topo = init_topology(dp, pp, backend=args.backend, device=device,
+                         tp=tp)
     device = topo.device
     is_gpu = device.type == "cuda"
 
@@ -79,7 +89,13 @@ def main():
     global_batch = args.local_batch * dp
     mubatch = args.local_batch // args.mubatches
 
-    model = MLP(sizes, topo.stage_id, pp, global_batch, loss=args.loss)
+    if tp > 1:
+        from shallowspeed_amd.parallel import TPMLP
+
+        model = TPMLP(sizes, topo.tp_group, topo.tp_rank, tp, global_batch,
+                      loss=args.loss)
+    else:
+        model = MLP(sizes, topo.stage_id, pp, global_batch, loss=args.loss)
     model.materialize_device(device)
     if args.fp8_fwd:
         model.set_fp8_fwd(True)
@@ -160,7 +176,9 @@ def main():
                                         + ("..." if len(sizes) > 4 else ""))),
                 "global_batch": global_batch,
                 "seq_len": None,
-                "parallelism": f"dp{dp}" + (f"xpp{pp}" if pp > 1 else ""),
+                "parallelism": f"dp{dp}"
+                + (f"xpp{pp}" if pp > 1 else "")
+                + (f"xtp{tp}" if tp > 1 else ""),
                 "schedule": args.schedule,
                 "loss": args.loss,
                 "mubatches": args.mubatches,

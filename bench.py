@@ -8,7 +8,10 @@ node as DP over RCCL/xGMI (weak scaling: fixed per-GPU batch).
 For N>1 the driver launches this via torch.distributed.run with one
 rank per GPU; RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* come from the env.
 
-Rank 0 prints ONE JSON line with the whole-job samples/sec.
+Rank 0 prints ONE JSON line with the whole-job samples/sec.  (Under
+the production nccl/RCCL backend stdout carries only that line; the
+gloo TEST backend prints its own "[Gloo] Rank..." connectivity banners
+to stdout — pipe through `grep '^{'` when parsing gloo runs.)
 """
 
 import argparse
